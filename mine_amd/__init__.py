@@ -1,0 +1,27 @@
+"""mine_amd — MI355X-native continuous-depth-MPI framework.
+
+A from-scratch implementation of the capabilities of vincentfung13/MINE
+("MINE: Towards Continuous Depth MPI with NeRF for Novel View Synthesis",
+ICCV 2021), designed MI355X-first:
+
+  * PyTorch-ROCm frontend (autograd, optimizer, data loading)
+  * hand-written HIP/CDNA4 kernels (gfx950) for the hot rendering path:
+    fused plane-sweep homography warp + z-cull + over-composite across all
+    N depth planes, fused src-view volume compositing with RGB blending,
+    fused SSIM — no per-plane intermediates are ever materialized
+  * RCCL over xGMI for data-parallel training (own bucketed gradient
+    all-reduce engine overlapped with backward)
+
+Layer map mirrors SURVEY.md section 1 of the reference analysis:
+  config   — mine_amd.config        (ref: train.py:30-56)
+  engine   — mine_amd.engine        (ref: synthesis_task.py)
+  models   — mine_amd.models        (ref: network/)
+  ops      — mine_amd.ops           (ref: operations/)
+  data     — mine_amd.data          (ref: input_pipelines/)
+  parallel — mine_amd.parallel      (ref: DDP/SyncBN usage)
+  utils    — mine_amd.utils
+"""
+
+__version__ = "0.1.0"
+
+from mine_amd.config import Config, load_config  # noqa: F401

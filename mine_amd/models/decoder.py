@@ -1,0 +1,163 @@
+"""MPI decoder — UNet over encoder taps, conditioned on continuous disparity.
+
+Behavioral contract (ref network/monodepth2/depth_decoder.py:35-148):
+  * the BxS disparity values are positionally encoded (21 ch for
+    multires=10) and concatenated as constant-per-plane channels onto
+    EVERY encoder feature map, which is expanded B -> B*S — the
+    "continuous depth" mechanism;
+  * a receptive-field neck (maxpool + 1x1 2048->512, maxpool + 3x3
+    512->256, up + 3x3 256->256, up + 1x1 256->2048, each BN+LeakyReLU(0.1))
+    runs at batch B before the expansion;
+  * 5 up-stages of [ConvBlock -> nearest x2 -> skip concat -> ConvBlock]
+    with channels [256,128,64,32,16], where ConvBlock = reflection-pad
+    3x3 conv + BN + ELU (ref network/monodepth2/layers.py:106-138);
+  * per-scale `dispconv` 3x3 -> 4 channels split into sigmoid RGB and
+    abs(x)+1e-4 sigma (or sigmoid alpha), with optional sigma dropout;
+    outputs {("disp", s): BxSx4xH_sxW_s} for s in scales.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Sequence, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from mine_amd.utils.embedder import PositionalEncoder
+
+
+class ConvBlock(nn.Module):
+    """ReflectionPad(1) + 3x3 conv + BN + ELU (ref monodepth2/layers.py:106-138)."""
+
+    def __init__(self, in_ch: int, out_ch: int):
+        super().__init__()
+        self.pad = nn.ReflectionPad2d(1)
+        self.conv = nn.Conv2d(in_ch, out_ch, 3)
+        self.bn = nn.BatchNorm2d(out_ch)
+        self.act = nn.ELU(inplace=True)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.act(self.bn(self.conv(self.pad(x))))
+
+
+def _neck_conv(in_ch: int, out_ch: int, k: int) -> nn.Sequential:
+    return nn.Sequential(
+        nn.Conv2d(in_ch, out_ch, k, stride=1, padding=(k - 1) // 2, bias=False),
+        nn.BatchNorm2d(out_ch),
+        nn.LeakyReLU(0.1, inplace=True),
+    )
+
+
+class Conv3x3Refl(nn.Module):
+    """ReflectionPad(1) + plain 3x3 conv (the dispconv head)."""
+
+    def __init__(self, in_ch: int, out_ch: int):
+        super().__init__()
+        self.pad = nn.ReflectionPad2d(1)
+        self.conv = nn.Conv2d(in_ch, out_ch, 3)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.conv(self.pad(x))
+
+
+class MPIDecoder(nn.Module):
+    def __init__(self,
+                 num_ch_enc: Sequence[int],
+                 pos_encoding_multires: int = 10,
+                 use_alpha: bool = False,
+                 num_output_channels: int = 4,
+                 scales: Sequence[int] = (0, 1, 2, 3),
+                 use_skips: bool = True,
+                 sigma_dropout_rate: float = 0.0):
+        super().__init__()
+        self.use_alpha = use_alpha
+        self.use_skips = use_skips
+        self.scales = list(scales)
+        self.num_output_channels = num_output_channels
+        self.sigma_dropout_rate = sigma_dropout_rate
+
+        self.embedder = PositionalEncoder(pos_encoding_multires, input_dims=1)
+        E = self.embedder.out_dim
+        self.E = E
+
+        c_last = num_ch_enc[-1]
+        self.downsample = nn.MaxPool2d(3, stride=2, padding=1)
+        self.upsample = nn.UpsamplingNearest2d(scale_factor=2)
+        self.conv_down1 = _neck_conv(c_last, 512, 1)
+        self.conv_down2 = _neck_conv(512, 256, 3)
+        self.conv_up1 = _neck_conv(256, 256, 3)
+        self.conv_up2 = _neck_conv(256, c_last, 1)
+
+        # channels after PE concat
+        ch_enc = [c + E for c in num_ch_enc]
+        ch_dec = [16, 32, 64, 128, 256]
+
+        self.upconvs0 = nn.ModuleList()
+        self.upconvs1 = nn.ModuleList()
+        for i in range(4, -1, -1):
+            c_in = ch_enc[-1] if i == 4 else ch_dec[i + 1]
+            self.upconvs0.append(ConvBlock(c_in, ch_dec[i]))
+            c_in = ch_dec[i]
+            if use_skips and i > 0:
+                c_in += ch_enc[i - 1]
+            self.upconvs1.append(ConvBlock(c_in, ch_dec[i]))
+        self.dispconvs = nn.ModuleDict({
+            str(s): Conv3x3Refl(ch_dec[s], num_output_channels) for s in self.scales
+        })
+
+    def _expand_with_pe(self, feat: torch.Tensor, pe: torch.Tensor,
+                        B: int, S: int) -> torch.Tensor:
+        """feat: BxCxHxW -> (B*S)x(C+E)xHxW with the plane's PE broadcast
+        over HxW (ref depth_decoder.py:103-116)."""
+        _, C, H, W = feat.shape
+        f = feat.unsqueeze(1).expand(B, S, C, H, W).reshape(B * S, C, H, W)
+        p = pe.to(feat.dtype).unsqueeze(-1).unsqueeze(-1).expand(B * S, self.E, H, W)
+        return torch.cat((f, p), dim=1)
+
+    def forward(self, input_features: List[torch.Tensor],
+                disparity: torch.Tensor) -> Dict[Tuple[str, int], torch.Tensor]:
+        """input_features: 5 taps BxCxHxW; disparity: BxS.
+
+        Returns {("disp", s): BxSx4xH_sxW_s} with rgb=sigmoid, sigma=|x|+1e-4.
+        """
+        B, S = disparity.shape
+        pe = self.embedder(disparity.reshape(B * S, 1))  # (B*S, E)
+
+        # receptive-field neck at batch B
+        enc_out = input_features[-1]
+        x = self.conv_down1(self.downsample(enc_out))
+        x = self.conv_down2(self.downsample(x))
+        x = self.conv_up1(self.upsample(x))
+        neck = self.conv_up2(self.upsample(x))  # BxC_lastxH/32xW/32
+
+        skips = [self._expand_with_pe(f, pe, B, S) for f in input_features]
+        x = self._expand_with_pe(neck, pe, B, S)
+
+        outputs: Dict[Tuple[str, int], torch.Tensor] = {}
+        for idx, i in enumerate(range(4, -1, -1)):
+            x = self.upconvs0[idx](x)
+            x = self.upsample(x)
+            if self.use_skips and i > 0:
+                skip = skips[i - 1]
+                if x.shape[-2:] != skip.shape[-2:]:
+                    # odd intermediate sizes (e.g. H/32 == 3): align to the
+                    # skip tap (the reference decoder requires power-of-two
+                    # -divisible sizes and would crash here)
+                    x = F.interpolate(x, size=skip.shape[-2:], mode="nearest")
+                x = torch.cat((x, skip), dim=1)
+            x = self.upconvs1[idx](x)
+            if i in self.scales:
+                out = self.dispconvs[str(i)](x)
+                Hs, Ws = out.shape[-2:]
+                mpi = out.view(B, S, self.num_output_channels, Hs, Ws)
+                rgb = torch.sigmoid(mpi[:, :, 0:3])
+                if self.use_alpha:
+                    sigma = torch.sigmoid(mpi[:, :, 3:])
+                else:
+                    sigma = torch.abs(mpi[:, :, 3:]) + 1e-4
+                if self.sigma_dropout_rate > 0.0 and self.training:
+                    sigma = F.dropout2d(
+                        sigma.view(B * S, 1, Hs, Ws), p=self.sigma_dropout_rate
+                    ).view(B, S, 1, Hs, Ws)
+                outputs[("disp", i)] = torch.cat((rgb, sigma), dim=2)
+        return outputs

@@ -1,0 +1,129 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark — RealEstate10K 384x256 N=64, bf16, DDP.
+
+Measures the BASELINE.json headline metric: train imgs/sec (whole job)
+on the reference's flagship config (384x256, 64 planes, per-GPU batch 4,
+ref configs/params_realestate.yaml) with synthetic data of that shape and
+random-init weights (no network access in this environment).
+
+    python bench.py --gpus 1 --steps 30 --warmup 10
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+        --master-addr 127.0.0.1 bench.py --gpus 8 --steps 30 --warmup 10
+
+One rank per GPU over RCCL; rank 0 prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=4, help="per-GPU batch size")
+    p.add_argument("--planes", type=int, default=64)
+    p.add_argument("--height", type=int, default=256)
+    p.add_argument("--width", type=int, default=384)
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--dataset", type=str, default="realestate10k")
+    args = p.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    from mine_amd.config import RuntimeState, default_config
+    from mine_amd.data import SyntheticMPIDataset, collate_src_tgt
+    from mine_amd.engine import SynthesisTask
+    from mine_amd.parallel import init_distributed
+
+    rank, local_rank, world_size = init_distributed()
+    if world_size > 1:
+        assert world_size == args.gpus, (world_size, args.gpus)
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
+    device = f"cuda:{local_rank % torch.cuda.device_count()}" if use_gpu else "cpu"
+
+    cfg = default_config(**{
+        "data.name": args.dataset,
+        "data.img_h": args.height, "data.img_w": args.width,
+        "mpi.num_bins_coarse": args.planes,
+        "data.per_gpu_batch_size": args.batch,
+        "data.visible_point_count": 256,
+        "lr.backbone_lr": 0.0002, "lr.decay_steps": [4, 8],
+        "training.amp_dtype": args.dtype,
+    })
+    state = RuntimeState(global_rank=rank, local_rank=local_rank,
+                         world_size=world_size)
+    task = SynthesisTask(cfg, state=state)
+
+    # Pre-build a few host-side batches; the H2D staging stays inside the
+    # timed region (it is part of a real training step).
+    ds = SyntheticMPIDataset(cfg, length=args.batch * 4)
+    batches = [collate_src_tgt([ds[i * args.batch + j] for j in range(args.batch)])
+               for i in range(4)]
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        if world_size > 1:
+            dist.barrier()
+            if use_gpu:
+                torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        task.train_step(batches[i % len(batches)])
+
+    sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        task.train_step(batches[i % len(batches)])
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks = whole-job wall time
+    if world_size > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    imgs_per_sec = world_size * args.batch * args.steps / elapsed
+    if rank == 0:
+        result = {
+            "metric": "train imgs/sec (whole node) RealEstate10K 384x256 N=64",
+            "value": round(imgs_per_sec, 3),
+            "unit": "imgs/sec",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "monodepth2-resnet50-mpi",
+                "global_batch": world_size * args.batch,
+                "seq_len": args.planes,
+                "parallelism": f"dp{world_size}",
+                "img_h": args.height,
+                "img_w": args.width,
+                "n_planes": args.planes,
+            },
+        }
+        print(json.dumps(result))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,181 @@
+// Torch bindings for the mine_amd HIP kernels (gfx950 only).
+//
+// Tensor-level contracts are documented in mine_amd/ops/renderer.py and
+// mine_amd/ops/ssim.py. Everything here is thin glue: shape checks,
+// output allocation, stream plumbing.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+extern "C" {
+void mine_src_composite_fwd(const float*, const float*, const float*,
+                            const float*, float*, float*, float*, int, int,
+                            int, int, int, hipStream_t);
+void mine_src_composite_bwd(const float*, const float*, const float*,
+                            const float*, const float*, const float*,
+                            const float*, float*, int, int, int, int, int,
+                            hipStream_t);
+void mine_tgt_composite_fwd(const float*, const float*, const float*,
+                            const float*, const float*, float*, float*,
+                            float*, int, int, int, int, int, hipStream_t);
+void mine_tgt_composite_bwd(const float*, const float*, const float*,
+                            const float*, const float*, const float*,
+                            const float*, float*, int, int, int, int, int,
+                            hipStream_t);
+void mine_ssim_set_window(const float*);
+void mine_ssim_fwd(const float*, const float*, float*, int, int, int,
+                   hipStream_t);
+void mine_ssim_bwd(const float*, const float*, const float*, float*, float*,
+                   float*, float*, int, int, int, hipStream_t);
+}
+
+namespace {
+
+#define CHECK_IN(x)                                               \
+  TORCH_CHECK((x).is_cuda(), #x " must be on GPU");               \
+  TORCH_CHECK((x).is_contiguous(), #x " must be contiguous");     \
+  TORCH_CHECK((x).scalar_type() == at::kFloat, #x " must be f32")
+
+const float* optr(const at::Tensor& t) {
+  return t.numel() ? t.data_ptr<float>() : nullptr;
+}
+
+hipStream_t stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+std::vector<at::Tensor> src_composite_fwd(at::Tensor mpi, at::Tensor depths,
+                                          at::Tensor kinv, at::Tensor img,
+                                          bool bg_inf) {
+  CHECK_IN(mpi); CHECK_IN(depths); CHECK_IN(kinv);
+  TORCH_CHECK(mpi.dim() == 5 && mpi.size(4) == 4, "mpi must be (B,S,H,W,4)");
+  const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
+  TORCH_CHECK(S <= 192, "S too large for the LDS geometry stage");
+  const bool blend = img.numel() > 0;
+  if (blend) {
+    CHECK_IN(img);
+    TORCH_CHECK(img.sizes() == at::IntArrayRef({B, H, W, 3}),
+                "img must be (B,H,W,3)");
+  }
+  auto rgb = at::empty({B, 3, H, W}, mpi.options());
+  auto depth = at::empty({B, 1, H, W}, mpi.options());
+  auto mpi_blend = blend ? at::empty_like(mpi) : at::empty({0}, mpi.options());
+  mine_src_composite_fwd(mpi.data_ptr<float>(), depths.data_ptr<float>(),
+                         kinv.data_ptr<float>(), optr(img),
+                         rgb.data_ptr<float>(), depth.data_ptr<float>(),
+                         blend ? mpi_blend.data_ptr<float>() : nullptr,
+                         B, S, H, W, bg_inf ? 1 : 0, stream());
+  return {rgb, depth, mpi_blend};
+}
+
+at::Tensor src_composite_bwd(at::Tensor mpi, at::Tensor depths,
+                             at::Tensor kinv, at::Tensor img, bool bg_inf,
+                             at::Tensor g_rgb, at::Tensor g_depth,
+                             at::Tensor g_blend) {
+  CHECK_IN(mpi);
+  const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
+  auto grad_mpi = at::empty_like(mpi);
+  mine_src_composite_bwd(mpi.data_ptr<float>(), depths.data_ptr<float>(),
+                         kinv.data_ptr<float>(), optr(img), optr(g_rgb),
+                         optr(g_depth), optr(g_blend),
+                         grad_mpi.data_ptr<float>(), B, S, H, W,
+                         bg_inf ? 1 : 0, stream());
+  return grad_mpi;
+}
+
+std::vector<at::Tensor> tgt_composite_fwd(at::Tensor mpi, at::Tensor hinv,
+                                          at::Tensor m, at::Tensor tvec,
+                                          at::Tensor depths, bool bg_inf) {
+  CHECK_IN(mpi); CHECK_IN(hinv); CHECK_IN(m); CHECK_IN(tvec); CHECK_IN(depths);
+  TORCH_CHECK(mpi.dim() == 5 && mpi.size(4) == 4, "mpi must be (B,S,H,W,4)");
+  const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
+  TORCH_CHECK(S <= 192, "S too large for the LDS geometry stage");
+  TORCH_CHECK(hinv.sizes() == at::IntArrayRef({B, S, 3, 3}));
+  auto rgb = at::empty({B, 3, H, W}, mpi.options());
+  auto depth = at::empty({B, 1, H, W}, mpi.options());
+  auto mask = at::empty({B, 1, H, W}, mpi.options());
+  mine_tgt_composite_fwd(mpi.data_ptr<float>(), hinv.data_ptr<float>(),
+                         m.data_ptr<float>(), tvec.data_ptr<float>(),
+                         depths.data_ptr<float>(), rgb.data_ptr<float>(),
+                         depth.data_ptr<float>(), mask.data_ptr<float>(),
+                         B, S, H, W, bg_inf ? 1 : 0, stream());
+  return {rgb, depth, mask};
+}
+
+at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor m,
+                             at::Tensor tvec, at::Tensor depths, bool bg_inf,
+                             at::Tensor g_rgb, at::Tensor g_depth) {
+  CHECK_IN(mpi);
+  const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
+  auto grad_mpi = at::zeros_like(mpi);
+  mine_tgt_composite_bwd(mpi.data_ptr<float>(), hinv.data_ptr<float>(),
+                         m.data_ptr<float>(), tvec.data_ptr<float>(),
+                         depths.data_ptr<float>(), optr(g_rgb), optr(g_depth),
+                         grad_mpi.data_ptr<float>(), B, S, H, W,
+                         bg_inf ? 1 : 0, stream());
+  return grad_mpi;
+}
+
+// --------------------------------------------------------------------------
+
+bool g_window_set = false;
+
+void ensure_window() {
+  if (g_window_set) return;
+  // Gaussian(11, sigma=1.5), normalized (ref network/ssim.py:7-9)
+  double w[11], sum = 0.0;
+  for (int i = 0; i < 11; ++i) {
+    const double d = i - 5;
+    w[i] = std::exp(-d * d / (2.0 * 1.5 * 1.5));
+    sum += w[i];
+  }
+  float wf[11];
+  for (int i = 0; i < 11; ++i) wf[i] = (float)(w[i] / sum);
+  mine_ssim_set_window(wf);
+  g_window_set = true;
+}
+
+std::vector<at::Tensor> ssim_fwd(at::Tensor img1, at::Tensor img2) {
+  CHECK_IN(img1); CHECK_IN(img2);
+  TORCH_CHECK(img1.dim() == 4 && img1.sizes() == img2.sizes());
+  ensure_window();
+  const int BC = img1.size(0) * img1.size(1);
+  const int H = img1.size(2), W = img1.size(3);
+  auto sum = at::zeros({1}, img1.options());
+  mine_ssim_fwd(img1.data_ptr<float>(), img2.data_ptr<float>(),
+                sum.data_ptr<float>(), BC, H, W, stream());
+  return {(sum / (double)(BC * (int64_t)H * W)).squeeze(0)};
+}
+
+at::Tensor ssim_bwd(at::Tensor img1, at::Tensor img2, at::Tensor gscale) {
+  CHECK_IN(img1); CHECK_IN(img2);
+  ensure_window();
+  const int BC = img1.size(0) * img1.size(1);
+  const int H = img1.size(2), W = img1.size(3);
+  auto F1 = at::empty_like(img1);
+  auto F2 = at::empty_like(img1);
+  auto F3 = at::empty_like(img1);
+  auto grad1 = at::empty_like(img1);
+  auto g = gscale.to(img1.options()).contiguous();
+  mine_ssim_bwd(img1.data_ptr<float>(), img2.data_ptr<float>(),
+                g.data_ptr<float>(), F1.data_ptr<float>(),
+                F2.data_ptr<float>(), F3.data_ptr<float>(),
+                grad1.data_ptr<float>(), BC, H, W, stream());
+  return grad1;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("src_composite_fwd", &src_composite_fwd,
+          "fused src-view MPI composite + RGB blend, forward");
+  mod.def("src_composite_bwd", &src_composite_bwd);
+  mod.def("tgt_composite_fwd", &tgt_composite_fwd,
+          "fused plane-sweep warp + z-cull + composite, forward");
+  mod.def("tgt_composite_bwd", &tgt_composite_bwd);
+  mod.def("ssim_fwd", &ssim_fwd);
+  mod.def("ssim_bwd", &ssim_bwd);
+}

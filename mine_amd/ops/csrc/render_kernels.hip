@@ -1,0 +1,636 @@
+// CDNA4 (gfx950) fused MPI rendering kernels.
+//
+// Design notes (MI355X-first, see /root/repo/SURVEY.md section 2b):
+//  * The MPI is packed (B, S, H, W, 4) = rgb+sigma adjacent per pixel, so
+//    every per-plane access is ONE naturally-aligned float4 — a wave's 64
+//    lanes walk x-adjacent pixels, giving 16 B/lane coalesced streams.
+//  * Per output pixel, ONE thread walks all S planes in registers: the
+//    compositing scan (shifted cumprod of transparencies) is a sequential
+//    recurrence per pixel but embarrassingly parallel across the B*H*W
+//    pixels (>3M at the flagship config — far above the 256-CU fill
+//    requirement of ~16k threads).
+//  * The novel-view kernel fuses homography projection, border-clamped
+//    bilinear sampling, analytic warped plane points (bilinear sampling of
+//    a linear field == evaluation at the mapped point), z-culling and the
+//    volume composite. The reference materialized nine BxSx{3,7}xHxW
+//    tensors per scale for this (ref operations/mpi_rendering.py:181-241 +
+//    homography_sampler.py:58-141); here nothing but the output leaves
+//    registers.
+//  * Backward recomputes the forward chain in three ascending passes per
+//    pixel (no per-plane state is saved): pass 1 accumulates the composite
+//    totals, pass 2 accumulates the cumprod-suffix total, pass 3 emits
+//    gradients. Ascending recomputation keeps the running transmittance
+//    product stable (no division by ~1e-6 factors).
+//  * Per-plane homographies/depths are staged in LDS per workgroup.
+//
+// Shapes: mpi (B,S,H,W,4) f32; images packed (B,H,W,3); composited outputs
+// (B,3,H,W) / (B,1,H,W) channel-major (what the loss stack consumes).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define DEV __device__ __forceinline__
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kMaxS = 192;  // LDS budget: 192 planes * 10 f32 = 7.5 KB
+
+struct float3x3 {
+  float m[9];
+  DEV float3 mul(float x, float y, float z) const {
+    return make_float3(m[0] * x + m[1] * y + m[2] * z,
+                       m[3] * x + m[4] * y + m[5] * z,
+                       m[6] * x + m[7] * y + m[8] * z);
+  }
+};
+
+DEV float clampf(float v, float lo, float hi) {
+  return fminf(fmaxf(v, lo), hi);
+}
+
+// ---------------------------------------------------------------------------
+// Source-view composite (+ optional RGB blending)
+// ---------------------------------------------------------------------------
+//   delta_s = |K^-1 p| * (d_{s+1} - d_s), far plane 1e3
+//   t = exp(-sigma*delta); u = t + 1e-6; A_s = prod_{j<s} u_j
+//   w = A * (1 - t); blended c = A*I + (1-A)*rgb
+//   R = sum w*c ; D = sum(w*z)/(sum w + 1e-5)  (or + 1000*(1-sum w))
+// ref operations/mpi_rendering.py:42-82, synthesis_task.py:267-274.
+
+template <bool BLEND, bool BG_INF>
+__global__ void __launch_bounds__(kBlock)
+src_composite_fwd_kernel(const float* __restrict__ mpi,
+                         const float* __restrict__ depths,   // (B,S)
+                         const float* __restrict__ kinv,     // (B,3,3)
+                         const float* __restrict__ img,      // (B,H,W,3)
+                         float* __restrict__ rgb_out,        // (B,3,H,W)
+                         float* __restrict__ depth_out,      // (B,1,H,W)
+                         float* __restrict__ mpi_blend,      // (B,S,H,W,4)
+                         int B, int S, int H, int W) {
+  __shared__ float s_depth[kMaxS];
+  const int b = blockIdx.y;
+  for (int i = threadIdx.x; i < S; i += kBlock) s_depth[i] = depths[b * S + i];
+  __syncthreads();
+
+  float3x3 Ki;
+#pragma unroll
+  for (int i = 0; i < 9; ++i) Ki.m[i] = kinv[b * 9 + i];
+
+  const int HW = H * W;
+  const int64_t mpi_b = (int64_t)b * S * HW * 4;
+  for (int pix = blockIdx.x * kBlock + threadIdx.x; pix < HW;
+       pix += gridDim.x * kBlock) {
+    const int y = pix / W;
+    const int x = pix - y * W;
+    const float3 ray = Ki.mul((float)x, (float)y, 1.0f);
+    const float nu = sqrtf(ray.x * ray.x + ray.y * ray.y + ray.z * ray.z);
+
+    float3 I = make_float3(0.f, 0.f, 0.f);
+    if (BLEND) {
+      const float* ip = img + ((int64_t)b * HW + pix) * 3;
+      I = make_float3(ip[0], ip[1], ip[2]);
+    }
+
+    float A = 1.0f, Wsum = 0.0f, Nsum = 0.0f;
+    float3 R = make_float3(0.f, 0.f, 0.f);
+    for (int s = 0; s < S; ++s) {
+      const float4 px = *reinterpret_cast<const float4*>(
+          mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
+      const float d = s_depth[s];
+      const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
+      const float t = __expf(-px.w * delta);
+      const float w = A * (1.0f - t);
+      float3 c = make_float3(px.x, px.y, px.z);
+      if (BLEND) {
+        c.x = A * I.x + (1.0f - A) * c.x;
+        c.y = A * I.y + (1.0f - A) * c.y;
+        c.z = A * I.z + (1.0f - A) * c.z;
+        float4 ob = make_float4(c.x, c.y, c.z, px.w);
+        *reinterpret_cast<float4*>(mpi_blend + mpi_b + ((int64_t)s * HW + pix) * 4) = ob;
+      }
+      R.x += w * c.x; R.y += w * c.y; R.z += w * c.z;
+      Wsum += w;
+      Nsum += w * d;  // src-view z of plane s is its depth
+      A *= (t + 1e-6f);
+    }
+    const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
+                           : (Nsum / (Wsum + 1e-5f));
+    rgb_out[((int64_t)b * 3 + 0) * HW + pix] = R.x;
+    rgb_out[((int64_t)b * 3 + 1) * HW + pix] = R.y;
+    rgb_out[((int64_t)b * 3 + 2) * HW + pix] = R.z;
+    depth_out[(int64_t)b * HW + pix] = D;
+  }
+}
+
+template <bool BLEND, bool BG_INF>
+__global__ void __launch_bounds__(kBlock)
+src_composite_bwd_kernel(const float* __restrict__ mpi,
+                         const float* __restrict__ depths,
+                         const float* __restrict__ kinv,
+                         const float* __restrict__ img,
+                         const float* __restrict__ g_rgb,    // (B,3,H,W) or null
+                         const float* __restrict__ g_depth,  // (B,1,H,W) or null
+                         const float* __restrict__ g_blend,  // (B,S,H,W,4) or null
+                         float* __restrict__ grad_mpi,       // (B,S,H,W,4)
+                         int B, int S, int H, int W) {
+  __shared__ float s_depth[kMaxS];
+  const int b = blockIdx.y;
+  for (int i = threadIdx.x; i < S; i += kBlock) s_depth[i] = depths[b * S + i];
+  __syncthreads();
+
+  float3x3 Ki;
+#pragma unroll
+  for (int i = 0; i < 9; ++i) Ki.m[i] = kinv[b * 9 + i];
+
+  const int HW = H * W;
+  const int64_t mpi_b = (int64_t)b * S * HW * 4;
+  for (int pix = blockIdx.x * kBlock + threadIdx.x; pix < HW;
+       pix += gridDim.x * kBlock) {
+    const int y = pix / W;
+    const int x = pix - y * W;
+    const float3 ray = Ki.mul((float)x, (float)y, 1.0f);
+    const float nu = sqrtf(ray.x * ray.x + ray.y * ray.y + ray.z * ray.z);
+
+    float3 I = make_float3(0.f, 0.f, 0.f);
+    if (BLEND) {
+      const float* ip = img + ((int64_t)b * HW + pix) * 3;
+      I = make_float3(ip[0], ip[1], ip[2]);
+    }
+    float3 gR = make_float3(0.f, 0.f, 0.f);
+    if (g_rgb) {
+      gR.x = g_rgb[((int64_t)b * 3 + 0) * HW + pix];
+      gR.y = g_rgb[((int64_t)b * 3 + 1) * HW + pix];
+      gR.z = g_rgb[((int64_t)b * 3 + 2) * HW + pix];
+    }
+    const float gD = g_depth ? g_depth[(int64_t)b * HW + pix] : 0.0f;
+
+    // ---- pass 1: composite totals ----
+    float A = 1.0f, Wsum = 0.0f, Nsum = 0.0f;
+    for (int s = 0; s < S; ++s) {
+      const float4 px = *reinterpret_cast<const float4*>(
+          mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
+      const float d = s_depth[s];
+      const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
+      const float t = __expf(-px.w * delta);
+      const float w = A * (1.0f - t);
+      Wsum += w;
+      Nsum += w * d;
+      A *= (t + 1e-6f);
+    }
+    const float Wp = Wsum + 1e-5f;
+    const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
+
+    // ---- pass 2: total cumprod-suffix mass ----
+    float TotalP = 0.0f;
+    A = 1.0f;
+    for (int s = 0; s < S; ++s) {
+      const float4 px = *reinterpret_cast<const float4*>(
+          mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
+      const float d = s_depth[s];
+      const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
+      const float t = __expf(-px.w * delta);
+      const float w = A * (1.0f - t);
+      float3 c = make_float3(px.x, px.y, px.z);
+      float3 gC = make_float3(0.f, 0.f, 0.f);
+      if (BLEND) {
+        c.x = A * I.x + (1.0f - A) * c.x;
+        c.y = A * I.y + (1.0f - A) * c.y;
+        c.z = A * I.z + (1.0f - A) * c.z;
+        if (g_blend) {
+          const float4 gb = *reinterpret_cast<const float4*>(
+              g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
+          gC = make_float3(gb.x, gb.y, gb.z);
+        }
+      }
+      const float3 dc = make_float3(w * gR.x + gC.x, w * gR.y + gC.y,
+                                    w * gR.z + gC.z);
+      const float e = c.x * gR.x + c.y * gR.y + c.z * gR.z +
+                      gD * (BG_INF ? (d - 1000.0f) : (d - D) / Wp);
+      float dA = (1.0f - t) * e;
+      if (BLEND) {
+        dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
+      }
+      TotalP += dA * A;
+      A *= (t + 1e-6f);
+    }
+
+    // ---- pass 3: emit gradients ----
+    float prefix = 0.0f;
+    A = 1.0f;
+    for (int s = 0; s < S; ++s) {
+      const float4 px = *reinterpret_cast<const float4*>(
+          mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
+      const float d = s_depth[s];
+      const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
+      const float t = __expf(-px.w * delta);
+      const float u = t + 1e-6f;
+      const float w = A * (1.0f - t);
+      float3 c = make_float3(px.x, px.y, px.z);
+      float3 gC = make_float3(0.f, 0.f, 0.f);
+      float gCs = 0.0f;
+      if (BLEND) {
+        c.x = A * I.x + (1.0f - A) * c.x;
+        c.y = A * I.y + (1.0f - A) * c.y;
+        c.z = A * I.z + (1.0f - A) * c.z;
+        if (g_blend) {
+          const float4 gb = *reinterpret_cast<const float4*>(
+              g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
+          gC = make_float3(gb.x, gb.y, gb.z);
+          gCs = gb.w;
+        }
+      }
+      const float3 dc = make_float3(w * gR.x + gC.x, w * gR.y + gC.y,
+                                    w * gR.z + gC.z);
+      const float e = c.x * gR.x + c.y * gR.y + c.z * gR.z +
+                      gD * (BG_INF ? (d - 1000.0f) : (d - D) / Wp);
+      float dA = (1.0f - t) * e;
+      if (BLEND) {
+        dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
+      }
+      prefix += dA * A;
+      const float dt = -A * e + (TotalP - prefix) / u;
+      const float dsigma = dt * (-delta * t) + gCs;
+      float4 g;
+      if (BLEND) {
+        const float oneA = 1.0f - A;
+        g = make_float4(oneA * dc.x, oneA * dc.y, oneA * dc.z, dsigma);
+      } else {
+        g = make_float4(dc.x, dc.y, dc.z, dsigma);
+      }
+      *reinterpret_cast<float4*>(grad_mpi + mpi_b + ((int64_t)s * HW + pix) * 4) = g;
+      A *= u;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Novel-view render: homography warp + z-cull + composite, fully fused
+// ---------------------------------------------------------------------------
+// Per (b,s) geometry staged in LDS: Hinv (tgt pixel -> src homogeneous,
+// 9 f32) and plane depth. M = R_tgt_src * K_src_inv and t give the warped
+// plane point analytically: v_s(p) = M * qhat_s(p) * d_s + t, where
+// qhat_s = (clamped src pixel, 1). ref operations/homography_sampler.py +
+// mpi_rendering.py:181-241.
+
+struct TapRef {
+  int x0, x1, y0, y1;
+  float wx, wy;  // fractional weights toward x1 / y1
+};
+
+DEV TapRef make_tap(float u, float v, int W, int H) {
+  TapRef t;
+  const float uc = clampf(u, 0.0f, (float)(W - 1));
+  const float vc = clampf(v, 0.0f, (float)(H - 1));
+  const float fx = floorf(uc), fy = floorf(vc);
+  t.x0 = (int)fx;
+  t.y0 = (int)fy;
+  t.x1 = min(t.x0 + 1, W - 1);
+  t.y1 = min(t.y0 + 1, H - 1);
+  t.wx = uc - fx;
+  t.wy = vc - fy;
+  return t;
+}
+
+DEV float4 bilinear4(const float* base, const TapRef& t, int W) {
+  const float4 p00 = *reinterpret_cast<const float4*>(base + ((int64_t)t.y0 * W + t.x0) * 4);
+  const float4 p01 = *reinterpret_cast<const float4*>(base + ((int64_t)t.y0 * W + t.x1) * 4);
+  const float4 p10 = *reinterpret_cast<const float4*>(base + ((int64_t)t.y1 * W + t.x0) * 4);
+  const float4 p11 = *reinterpret_cast<const float4*>(base + ((int64_t)t.y1 * W + t.x1) * 4);
+  const float w00 = (1 - t.wx) * (1 - t.wy), w01 = t.wx * (1 - t.wy);
+  const float w10 = (1 - t.wx) * t.wy, w11 = t.wx * t.wy;
+  return make_float4(w00 * p00.x + w01 * p01.x + w10 * p10.x + w11 * p11.x,
+                     w00 * p00.y + w01 * p01.y + w10 * p10.y + w11 * p11.y,
+                     w00 * p00.z + w01 * p01.z + w10 * p10.z + w11 * p11.z,
+                     w00 * p00.w + w01 * p01.w + w10 * p10.w + w11 * p11.w);
+}
+
+// One plane's warped sample: rgb+sigma (z-culled), analytic point, validity.
+struct PlaneSample {
+  float4 rgbs;
+  float3 v;
+  float inb;
+};
+
+DEV PlaneSample sample_plane(const float* __restrict__ mpi_b, int s, int HW,
+                             const float* __restrict__ s_geom, float d,
+                             const float3x3& M, float3 tv,
+                             int x, int y, int W, int H, TapRef* tap_out) {
+  const float* Hrow = s_geom + s * 9;
+  const float hx = Hrow[0] * x + Hrow[1] * y + Hrow[2];
+  const float hy = Hrow[3] * x + Hrow[4] * y + Hrow[5];
+  const float hz = Hrow[6] * x + Hrow[7] * y + Hrow[8];
+  const float iz = 1.0f / hz;
+  const float u = hx * iz, v = hy * iz;
+
+  PlaneSample ps;
+  ps.inb = (u > -1.0f && u < (float)W && v > -1.0f && v < (float)H) ? 1.0f : 0.0f;
+
+  const TapRef tap = make_tap(u, v, W, H);
+  if (tap_out) *tap_out = tap;
+  ps.rgbs = bilinear4(mpi_b + (int64_t)s * HW * 4, tap, W);
+
+  const float ucl = clampf(u, 0.0f, (float)(W - 1));
+  const float vcl = clampf(v, 0.0f, (float)(H - 1));
+  const float3 r = M.mul(ucl, vcl, 1.0f);
+  ps.v = make_float3(r.x * d + tv.x, r.y * d + tv.y, r.z * d + tv.z);
+  if (ps.v.z < 0.0f) ps.rgbs.w = 0.0f;  // z-cull (ref mpi_rendering.py:233-235)
+  return ps;
+}
+
+template <bool BG_INF>
+__global__ void __launch_bounds__(kBlock)
+tgt_composite_fwd_kernel(const float* __restrict__ mpi,
+                         const float* __restrict__ hinv,    // (B,S,3,3)
+                         const float* __restrict__ m_rki,   // (B,3,3)
+                         const float* __restrict__ tvec,    // (B,3)
+                         const float* __restrict__ depths,  // (B,S)
+                         float* __restrict__ rgb_out,
+                         float* __restrict__ depth_out,
+                         float* __restrict__ mask_out,
+                         int B, int S, int H, int W) {
+  __shared__ float s_geom[kMaxS * 9];
+  __shared__ float s_depth[kMaxS];
+  const int b = blockIdx.y;
+  for (int i = threadIdx.x; i < S * 9; i += kBlock) s_geom[i] = hinv[(int64_t)b * S * 9 + i];
+  for (int i = threadIdx.x; i < S; i += kBlock) s_depth[i] = depths[b * S + i];
+  __syncthreads();
+
+  float3x3 M;
+#pragma unroll
+  for (int i = 0; i < 9; ++i) M.m[i] = m_rki[b * 9 + i];
+  const float3 tv = make_float3(tvec[b * 3], tvec[b * 3 + 1], tvec[b * 3 + 2]);
+
+  const int HW = H * W;
+  const float* mpi_b = mpi + (int64_t)b * S * HW * 4;
+  for (int pix = blockIdx.x * kBlock + threadIdx.x; pix < HW;
+       pix += gridDim.x * kBlock) {
+    const int y = pix / W;
+    const int x = pix - y * W;
+
+    float A = 1.0f, Wsum = 0.0f, Nsum = 0.0f, mask = 0.0f;
+    float3 R = make_float3(0.f, 0.f, 0.f);
+
+    PlaneSample cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv,
+                                   x, y, W, H, nullptr);
+    mask += cur.inb;
+    for (int s = 0; s < S; ++s) {
+      float delta;
+      PlaneSample nxt;
+      if (s + 1 < S) {
+        nxt = sample_plane(mpi_b, s + 1, HW, s_geom, s_depth[s + 1], M, tv,
+                           x, y, W, H, nullptr);
+        mask += nxt.inb;
+        const float dx = nxt.v.x - cur.v.x, dy = nxt.v.y - cur.v.y,
+                    dz = nxt.v.z - cur.v.z;
+        delta = sqrtf(dx * dx + dy * dy + dz * dz);
+      } else {
+        delta = 1e3f;
+      }
+      const float t = __expf(-cur.rgbs.w * delta);
+      const float w = A * (1.0f - t);
+      R.x += w * cur.rgbs.x; R.y += w * cur.rgbs.y; R.z += w * cur.rgbs.z;
+      Wsum += w;
+      Nsum += w * cur.v.z;
+      A *= (t + 1e-6f);
+      cur = nxt;
+    }
+    const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
+                           : (Nsum / (Wsum + 1e-5f));
+    rgb_out[((int64_t)b * 3 + 0) * HW + pix] = R.x;
+    rgb_out[((int64_t)b * 3 + 1) * HW + pix] = R.y;
+    rgb_out[((int64_t)b * 3 + 2) * HW + pix] = R.z;
+    depth_out[(int64_t)b * HW + pix] = D;
+    mask_out[(int64_t)b * HW + pix] = mask;
+  }
+}
+
+DEV void scatter4(float* base, const TapRef& t, int W, float4 g) {
+  const float w00 = (1 - t.wx) * (1 - t.wy), w01 = t.wx * (1 - t.wy);
+  const float w10 = (1 - t.wx) * t.wy, w11 = t.wx * t.wy;
+  float* p00 = base + ((int64_t)t.y0 * W + t.x0) * 4;
+  float* p01 = base + ((int64_t)t.y0 * W + t.x1) * 4;
+  float* p10 = base + ((int64_t)t.y1 * W + t.x0) * 4;
+  float* p11 = base + ((int64_t)t.y1 * W + t.x1) * 4;
+#define SC(P, WGT)                                  \
+  if (WGT != 0.0f) {                                \
+    atomicAdd(P + 0, WGT * g.x);                    \
+    atomicAdd(P + 1, WGT * g.y);                    \
+    atomicAdd(P + 2, WGT * g.z);                    \
+    atomicAdd(P + 3, WGT * g.w);                    \
+  }
+  SC(p00, w00) SC(p01, w01) SC(p10, w10) SC(p11, w11)
+#undef SC
+}
+
+template <bool BG_INF>
+__global__ void __launch_bounds__(kBlock)
+tgt_composite_bwd_kernel(const float* __restrict__ mpi,
+                         const float* __restrict__ hinv,
+                         const float* __restrict__ m_rki,
+                         const float* __restrict__ tvec,
+                         const float* __restrict__ depths,
+                         const float* __restrict__ g_rgb,
+                         const float* __restrict__ g_depth,
+                         float* __restrict__ grad_mpi,  // pre-zeroed
+                         int B, int S, int H, int W) {
+  __shared__ float s_geom[kMaxS * 9];
+  __shared__ float s_depth[kMaxS];
+  const int b = blockIdx.y;
+  for (int i = threadIdx.x; i < S * 9; i += kBlock) s_geom[i] = hinv[(int64_t)b * S * 9 + i];
+  for (int i = threadIdx.x; i < S; i += kBlock) s_depth[i] = depths[b * S + i];
+  __syncthreads();
+
+  float3x3 M;
+#pragma unroll
+  for (int i = 0; i < 9; ++i) M.m[i] = m_rki[b * 9 + i];
+  const float3 tv = make_float3(tvec[b * 3], tvec[b * 3 + 1], tvec[b * 3 + 2]);
+
+  const int HW = H * W;
+  const float* mpi_b = mpi + (int64_t)b * S * HW * 4;
+  float* gm_b = grad_mpi + (int64_t)b * S * HW * 4;
+  for (int pix = blockIdx.x * kBlock + threadIdx.x; pix < HW;
+       pix += gridDim.x * kBlock) {
+    const int y = pix / W;
+    const int x = pix - y * W;
+
+    float3 gR = make_float3(0.f, 0.f, 0.f);
+    if (g_rgb) {
+      gR.x = g_rgb[((int64_t)b * 3 + 0) * HW + pix];
+      gR.y = g_rgb[((int64_t)b * 3 + 1) * HW + pix];
+      gR.z = g_rgb[((int64_t)b * 3 + 2) * HW + pix];
+    }
+    const float gD = g_depth ? g_depth[(int64_t)b * HW + pix] : 0.0f;
+
+    // ---- pass 1 ----
+    float A = 1.0f, Wsum = 0.0f, Nsum = 0.0f;
+    PlaneSample cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv,
+                                   x, y, W, H, nullptr);
+    for (int s = 0; s < S; ++s) {
+      float delta;
+      PlaneSample nxt;
+      if (s + 1 < S) {
+        nxt = sample_plane(mpi_b, s + 1, HW, s_geom, s_depth[s + 1], M, tv,
+                           x, y, W, H, nullptr);
+        const float dx = nxt.v.x - cur.v.x, dy = nxt.v.y - cur.v.y,
+                    dz = nxt.v.z - cur.v.z;
+        delta = sqrtf(dx * dx + dy * dy + dz * dz);
+      } else {
+        delta = 1e3f;
+      }
+      const float t = __expf(-cur.rgbs.w * delta);
+      const float w = A * (1.0f - t);
+      Wsum += w;
+      Nsum += w * cur.v.z;
+      A *= (t + 1e-6f);
+      cur = nxt;
+    }
+    const float Wp = Wsum + 1e-5f;
+    const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
+
+    // ---- pass 2 ----
+    float TotalP = 0.0f;
+    A = 1.0f;
+    cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
+                       nullptr);
+    for (int s = 0; s < S; ++s) {
+      float delta;
+      PlaneSample nxt;
+      if (s + 1 < S) {
+        nxt = sample_plane(mpi_b, s + 1, HW, s_geom, s_depth[s + 1], M, tv,
+                           x, y, W, H, nullptr);
+        const float dx = nxt.v.x - cur.v.x, dy = nxt.v.y - cur.v.y,
+                    dz = nxt.v.z - cur.v.z;
+        delta = sqrtf(dx * dx + dy * dy + dz * dz);
+      } else {
+        delta = 1e3f;
+      }
+      const float t = __expf(-cur.rgbs.w * delta);
+      const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
+                      gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
+      TotalP += (1.0f - t) * e * A;
+      A *= (t + 1e-6f);
+      cur = nxt;
+    }
+
+    // ---- pass 3: emit gradients, bilinear scatter ----
+    float prefix = 0.0f;
+    A = 1.0f;
+    TapRef tap;
+    cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
+                       &tap);
+    for (int s = 0; s < S; ++s) {
+      float delta;
+      PlaneSample nxt;
+      TapRef ntap;
+      if (s + 1 < S) {
+        nxt = sample_plane(mpi_b, s + 1, HW, s_geom, s_depth[s + 1], M, tv,
+                           x, y, W, H, &ntap);
+        const float dx = nxt.v.x - cur.v.x, dy = nxt.v.y - cur.v.y,
+                    dz = nxt.v.z - cur.v.z;
+        delta = sqrtf(dx * dx + dy * dy + dz * dz);
+      } else {
+        delta = 1e3f;
+      }
+      const float t = __expf(-cur.rgbs.w * delta);
+      const float u = t + 1e-6f;
+      const float w = A * (1.0f - t);
+      const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
+                      gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
+      prefix += (1.0f - t) * e * A;
+      const float dt = -A * e + (TotalP - prefix) / u;
+      // culled sigma contributed nothing -> no gradient through it
+      const float dsigma = (cur.v.z < 0.0f) ? 0.0f : dt * (-delta * t);
+      scatter4(gm_b + (int64_t)s * HW * 4, tap, W,
+               make_float4(w * gR.x, w * gR.y, w * gR.z, dsigma));
+      A *= u;
+      cur = nxt;
+      tap = ntap;
+    }
+  }
+}
+
+inline int grid_x(int HW) {
+  int g = (HW + kBlock - 1) / kBlock;
+  return g < 4096 ? g : 4096;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+#define DISPATCH_BOOL(VAL, NAME, ...)          \
+  if (VAL) {                                   \
+    constexpr bool NAME = true;                \
+    __VA_ARGS__;                               \
+  } else {                                     \
+    constexpr bool NAME = false;               \
+    __VA_ARGS__;                               \
+  }
+
+extern "C" {
+
+void mine_src_composite_fwd(const float* mpi, const float* depths,
+                            const float* kinv, const float* img,
+                            float* rgb_out, float* depth_out, float* mpi_blend,
+                            int B, int S, int H, int W, int bg_inf,
+                            hipStream_t stream) {
+  dim3 grid(grid_x(H * W), B);
+  const bool blend = img != nullptr;
+  DISPATCH_BOOL(blend, BLEND, {
+    DISPATCH_BOOL(bg_inf, BG, {
+      hipLaunchKernelGGL((src_composite_fwd_kernel<BLEND, BG>), grid,
+                         dim3(kBlock), 0, stream, mpi, depths, kinv, img,
+                         rgb_out, depth_out, mpi_blend, B, S, H, W);
+    })
+  })
+}
+
+void mine_src_composite_bwd(const float* mpi, const float* depths,
+                            const float* kinv, const float* img,
+                            const float* g_rgb, const float* g_depth,
+                            const float* g_blend, float* grad_mpi,
+                            int B, int S, int H, int W, int bg_inf,
+                            hipStream_t stream) {
+  dim3 grid(grid_x(H * W), B);
+  const bool blend = img != nullptr;
+  DISPATCH_BOOL(blend, BLEND, {
+    DISPATCH_BOOL(bg_inf, BG, {
+      hipLaunchKernelGGL((src_composite_bwd_kernel<BLEND, BG>), grid,
+                         dim3(kBlock), 0, stream, mpi, depths, kinv, img,
+                         g_rgb, g_depth, g_blend, grad_mpi, B, S, H, W);
+    })
+  })
+}
+
+void mine_tgt_composite_fwd(const float* mpi, const float* hinv,
+                            const float* m_rki, const float* tvec,
+                            const float* depths, float* rgb_out,
+                            float* depth_out, float* mask_out,
+                            int B, int S, int H, int W, int bg_inf,
+                            hipStream_t stream) {
+  dim3 grid(grid_x(H * W), B);
+  DISPATCH_BOOL(bg_inf, BG, {
+    hipLaunchKernelGGL((tgt_composite_fwd_kernel<BG>), grid, dim3(kBlock), 0,
+                       stream, mpi, hinv, m_rki, tvec, depths, rgb_out,
+                       depth_out, mask_out, B, S, H, W);
+  })
+}
+
+void mine_tgt_composite_bwd(const float* mpi, const float* hinv,
+                            const float* m_rki, const float* tvec,
+                            const float* depths, const float* g_rgb,
+                            const float* g_depth, float* grad_mpi,
+                            int B, int S, int H, int W, int bg_inf,
+                            hipStream_t stream) {
+  dim3 grid(grid_x(H * W), B);
+  DISPATCH_BOOL(bg_inf, BG, {
+    hipLaunchKernelGGL((tgt_composite_bwd_kernel<BG>), grid, dim3(kBlock), 0,
+                       stream, mpi, hinv, m_rki, tvec, depths, g_rgb, g_depth,
+                       grad_mpi, B, S, H, W);
+  })
+}
+
+}  // extern "C"

@@ -126,7 +126,7 @@ def _bucket_boundaries(rank, world_size):
     assert total == sum(p.numel() for p in model.parameters())
     # grads are views into the buckets
     engine.zero_grad()
-    model(torch.randn(2, 1000)).sum().backward()
+    model(torch.randn(2, 256)).sum().backward()
     engine.finish_step()
     for b in engine.buckets:
         assert b.handle is None

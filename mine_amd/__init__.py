@@ -24,4 +24,13 @@ Layer map mirrors SURVEY.md section 1 of the reference analysis:
 
 __version__ = "0.1.0"
 
+import os as _os
+
+# MIOpen immediate/fast find: the default exhaustive per-shape kernel
+# search costs ~10 minutes of first-step time on a fresh box for this
+# model's ~40 conv shapes. FAST uses heuristic kernel selection (perf
+# within a few % for these implicit-GEMM shapes) and keeps cold-start in
+# seconds. Override by exporting MIOPEN_FIND_MODE before import.
+_os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 from mine_amd.config import Config, load_config  # noqa: F401

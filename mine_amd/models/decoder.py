@@ -21,6 +21,8 @@ from typing import Dict, List, Sequence, Tuple
 
 import torch
 import torch.nn as nn
+
+from mine_amd.models.norm import Fp32BatchNorm2d
 import torch.nn.functional as F
 
 from mine_amd.utils.embedder import PositionalEncoder
@@ -33,7 +35,7 @@ class ConvBlock(nn.Module):
         super().__init__()
         self.pad = nn.ReflectionPad2d(1)
         self.conv = nn.Conv2d(in_ch, out_ch, 3)
-        self.bn = nn.BatchNorm2d(out_ch)
+        self.bn = Fp32BatchNorm2d(out_ch)
         self.act = nn.ELU(inplace=True)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -43,7 +45,7 @@ class ConvBlock(nn.Module):
 def _neck_conv(in_ch: int, out_ch: int, k: int) -> nn.Sequential:
     return nn.Sequential(
         nn.Conv2d(in_ch, out_ch, k, stride=1, padding=(k - 1) // 2, bias=False),
-        nn.BatchNorm2d(out_ch),
+        Fp32BatchNorm2d(out_ch),
         nn.LeakyReLU(0.1, inplace=True),
     )
 

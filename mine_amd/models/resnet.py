@@ -24,6 +24,8 @@ from typing import List
 import torch
 import torch.nn as nn
 
+from mine_amd.models.norm import Fp32BatchNorm2d
+
 _IMAGENET_MEAN = (0.485, 0.456, 0.406)
 _IMAGENET_STD = (0.229, 0.224, 0.225)
 
@@ -36,11 +38,11 @@ class Bottleneck(nn.Module):
         super().__init__()
         out_ch = planes * self.expansion
         self.conv1 = nn.Conv2d(in_ch, planes, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = Fp32BatchNorm2d(planes)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = Fp32BatchNorm2d(planes)
         self.conv3 = nn.Conv2d(planes, out_ch, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(out_ch)
+        self.bn3 = Fp32BatchNorm2d(out_ch)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
 
@@ -64,7 +66,7 @@ class ResNetEncoder(nn.Module):
         self.num_ch_enc = [64, 256, 512, 1024, 2048]
 
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = Fp32BatchNorm2d(64)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
 
@@ -96,7 +98,7 @@ class ResNetEncoder(nn.Module):
         if stride != 1 or self.in_ch != out_ch:
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(out_ch),
+                Fp32BatchNorm2d(out_ch),
             )
         layers = [Bottleneck(self.in_ch, planes, stride, downsample)]
         self.in_ch = out_ch

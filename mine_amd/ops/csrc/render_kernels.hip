@@ -182,7 +182,10 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
 
     // ---- pass 2: total cumprod-suffix mass ----
-    float TotalP = 0.0f;
+    // fp64 accumulators: pass 3 computes (TotalP - prefix) / u with
+    // u ~ 1e-6; fp32 cancellation between O(1) terms would amplify to
+    // O(0.1) gradient error after the divide.
+    double TotalP = 0.0;
     A = 1.0f;
     for (int s = 0; s < S; ++s) {
       const float4 px = *reinterpret_cast<const float4*>(
@@ -211,12 +214,12 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
       if (BLEND) {
         dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
       }
-      TotalP += dA * A;
+      TotalP += (double)(dA * A);
       A *= (t + 1e-6f);
     }
 
     // ---- pass 3: emit gradients ----
-    float prefix = 0.0f;
+    double prefix = 0.0;
     A = 1.0f;
     for (int s = 0; s < S; ++s) {
       const float4 px = *reinterpret_cast<const float4*>(
@@ -248,8 +251,8 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
       if (BLEND) {
         dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
       }
-      prefix += dA * A;
-      const float dt = -A * e + (TotalP - prefix) / u;
+      prefix += (double)(dA * A);
+      const float dt = -A * e + (float)(TotalP - prefix) / u;
       const float dsigma = dt * (-delta * t) + gCs;
       float4 g;
       if (BLEND) {
@@ -488,8 +491,8 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
     const float Wp = Wsum + 1e-5f;
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
 
-    // ---- pass 2 ----
-    float TotalP = 0.0f;
+    // ---- pass 2 ---- (fp64 suffix accumulators; see src bwd note)
+    double TotalP = 0.0;
     A = 1.0f;
     cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
                        nullptr);
@@ -508,13 +511,13 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       const float t = __expf(-cur.rgbs.w * delta);
       const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
                       gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
-      TotalP += (1.0f - t) * e * A;
+      TotalP += (double)((1.0f - t) * e * A);
       A *= (t + 1e-6f);
       cur = nxt;
     }
 
     // ---- pass 3: emit gradients, bilinear scatter ----
-    float prefix = 0.0f;
+    double prefix = 0.0;
     A = 1.0f;
     TapRef tap;
     cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
@@ -537,8 +540,8 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       const float w = A * (1.0f - t);
       const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
                       gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
-      prefix += (1.0f - t) * e * A;
-      const float dt = -A * e + (TotalP - prefix) / u;
+      prefix += (double)((1.0f - t) * e * A);
+      const float dt = -A * e + (float)(TotalP - prefix) / u;
       // culled sigma contributed nothing -> no gradient through it
       const float dsigma = (cur.v.z < 0.0f) ? 0.0f : dt * (-delta * t);
       scatter4(gm_b + (int64_t)s * HW * 4, tap, W,

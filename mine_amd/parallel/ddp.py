@@ -126,10 +126,13 @@ class GradAllReduceEngine:
         flush()
 
         # Point param.grad at views into the flat buffers so accumulation
-        # writes land directly in the bucket.
+        # writes land directly in the bucket. The view copies the PARAM's
+        # strides (channels_last conv weights are NHWC-strided) so autograd's
+        # gradient layout contract holds and accumulation is a plain add.
         for b in self.buckets:
             for p, o in zip(b.params, b.offsets):
-                p.grad = b.flat[o:o + p.numel()].view_as(p)
+                flat = b.flat[o:o + p.numel()]
+                p.grad = flat.as_strided(p.shape, p.stride())
 
         self._hooks = []
         if self.enabled:

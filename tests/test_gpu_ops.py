@@ -80,8 +80,8 @@ def test_src_composite_backward_matches_oracle(bg_inf, blend):
                      generator=gseed)
 
     def run(device):
-        r = rgb.to(device).requires_grad_(True)
-        s = sigma.to(device).requires_grad_(True)
+        r = rgb.detach().clone().to(device).requires_grad_(True)
+        s = sigma.detach().clone().to(device).requires_grad_(True)
         mpi = pack_mpi(r, s)
         rgb_s, depth_s, blend_s = render_src_view(
             mpi, disparity.to(device), K_inv.to(device),
@@ -142,8 +142,8 @@ def test_tgt_composite_backward_matches_oracle(bg_inf):
     wd = torch.randn(rgb.shape[0], 1, rgb.shape[-2], rgb.shape[-1], generator=gseed)
 
     def run(device):
-        r = rgb.to(device).requires_grad_(True)
-        s = sigma.to(device).requires_grad_(True)
+        r = rgb.detach().clone().to(device).requires_grad_(True)
+        s = sigma.detach().clone().to(device).requires_grad_(True)
         if device == "cuda:0":
             mpi = pack_mpi(r, s)
             o_rgb, o_depth, _ = render_tgt_view(mpi, disparity.to(device),

@@ -1,0 +1,233 @@
+"""COLMAP I/O round-trip, LLFF dataset, LPIPS and video-CLI tests (CPU)."""
+import math
+import os
+import sys
+
+import numpy as np
+import pytest
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
+from mine_amd.data import colmap
+
+
+# ---------------------------------------------------------------------------
+# fixtures: a tiny synthetic COLMAP scene on disk
+# ---------------------------------------------------------------------------
+
+
+def _make_scene(root, n_views=4, n_points=80, W=64, H=48, ratio="7.875"):
+    """Write PNGs + a COLMAP .bin sparse model describing a toy scene."""
+    from PIL import Image as PILImage
+    rng = np.random.default_rng(7)
+    scene = os.path.join(root, "scene0")
+    sparse = os.path.join(scene, "sparse", "0")
+    img_dir = os.path.join(scene, f"images_{ratio}")
+    os.makedirs(sparse)
+    os.makedirs(img_dir)
+
+    f = 0.8 * W
+    cameras = {1: colmap.Camera(1, "SIMPLE_RADIAL", W, H,
+                                np.array([f, W / 2, H / 2, 0.0]))}
+
+    # world points in front of all cameras
+    pts_w = np.stack([rng.uniform(-1, 1, n_points),
+                      rng.uniform(-1, 1, n_points),
+                      rng.uniform(4.0, 10.0, n_points)], axis=0)
+
+    images = {}
+    points3d = {}
+    tracks = {pid: [] for pid in range(1, n_points + 1)}
+    for i in range(1, n_views + 1):
+        angle = 0.05 * (i - 1)
+        R = np.array([[math.cos(angle), 0, math.sin(angle)],
+                      [0, 1, 0],
+                      [-math.sin(angle), 0, math.cos(angle)]])
+        t = np.array([0.1 * (i - 1), 0.0, 0.0])
+        q = colmap.rotmat2qvec(R)
+        xyz_c = R @ pts_w + t[:, None]
+        uv = xyz_c[:2] / xyz_c[2:]
+        px = f * uv[0] + W / 2
+        py = f * uv[1] + H / 2
+        vis = (xyz_c[2] > 0.1) & (px >= 0) & (px < W) & (py >= 0) & (py < H)
+        pids = np.where(vis)[0] + 1
+        xys = np.stack([px[vis], py[vis]], axis=-1)
+        images[i] = colmap.Image(i, q, t, 1, f"view_{i:03d}.png",
+                                 xys, pids.astype(np.int64))
+        for k, pid in enumerate(pids):
+            tracks[int(pid)].append((i, k))
+        arr = (rng.uniform(0, 1, (H, W, 3)) * 255).astype(np.uint8)
+        PILImage.fromarray(arr).save(os.path.join(img_dir, f"view_{i:03d}.png"))
+
+    for pid in range(1, n_points + 1):
+        tr = tracks[pid]
+        points3d[pid] = colmap.Point3D(
+            pid, pts_w[:, pid - 1], np.array([128, 128, 128], dtype=np.uint8),
+            0.5, np.array([t[0] for t in tr], dtype=np.int32),
+            np.array([t[1] for t in tr], dtype=np.int32))
+
+    colmap.write_model(cameras, images, points3d, sparse)
+    return root
+
+
+# ---------------------------------------------------------------------------
+# COLMAP I/O
+# ---------------------------------------------------------------------------
+
+
+def test_qvec_rotmat_roundtrip():
+    rng = np.random.default_rng(0)
+    for _ in range(20):
+        q = rng.normal(size=4)
+        q /= np.linalg.norm(q)
+        if q[0] < 0:
+            q = -q
+        R = colmap.qvec2rotmat(q)
+        assert np.allclose(R @ R.T, np.eye(3), atol=1e-10)
+        assert np.isclose(np.linalg.det(R), 1.0)
+        q2 = colmap.rotmat2qvec(R)
+        assert np.allclose(q, q2, atol=1e-8)
+
+
+def test_colmap_binary_roundtrip(tmp_path):
+    _make_scene(str(tmp_path))
+    sparse = os.path.join(str(tmp_path), "scene0", "sparse", "0")
+    cams, imgs, pts = colmap.read_model(sparse, ".bin")
+    assert len(cams) == 1 and cams[1].model == "SIMPLE_RADIAL"
+    assert cams[1].width == 64 and cams[1].height == 48
+    K = cams[1].intrinsic_matrix()
+    assert K[0, 0] == K[1, 1] == pytest.approx(0.8 * 64)
+    assert len(imgs) == 4
+    for im in imgs.values():
+        assert im.xys.shape[0] == im.point3D_ids.shape[0] > 0
+        R = im.qvec2rotmat()
+        assert np.allclose(R @ R.T, np.eye(3), atol=1e-10)
+    assert len(pts) == 80
+    # geometric consistency: reproject a tracked point through its image
+    im = imgs[1]
+    pid = int(im.point3D_ids[0])
+    xyz_c = im.qvec2rotmat() @ pts[pid].xyz + im.tvec
+    uv = (cams[1].intrinsic_matrix() @ xyz_c)
+    uv = uv[:2] / uv[2]
+    assert np.allclose(uv, im.xys[0], atol=1e-6)
+
+
+# ---------------------------------------------------------------------------
+# LLFF dataset
+# ---------------------------------------------------------------------------
+
+
+def test_llff_dataset_items(tmp_path):
+    from mine_amd.config import default_config
+    from mine_amd.data.llff import NeRFDataset
+
+    _make_scene(str(tmp_path))
+    cfg = default_config(**{
+        "data.name": "llff", "data.img_h": 32, "data.img_w": 40,
+        "data.visible_point_count": 16,
+        "data.training_set_path": str(tmp_path)})
+    ds = NeRFDataset(cfg, root=str(tmp_path), img_size=(40, 32),
+                     visible_points_count=16)
+    assert len(ds) == 4
+    src, tgts = ds[0]
+    assert src["img"].shape == (3, 32, 40)
+    assert src["K"].shape == (3, 3) and src["K_inv"].shape == (3, 3)
+    assert torch.allclose(src["K"] @ src["K_inv"], torch.eye(3), atol=1e-5)
+    # K scaled to the 40x32 output from the 64x48 COLMAP camera
+    assert src["K"][0, 0].item() == pytest.approx(0.8 * 64 * 40 / 64)
+    assert src["xyzs"].shape == (3, 16)
+    assert (src["xyzs"][2] > 0).all()
+    assert len(tgts) == 1
+    t = tgts[0]
+    assert t["img"].shape == (3, 32, 40)
+    assert t["G_src_tgt"].shape == (4, 4)
+    # rigid: R orthonormal
+    R = t["G_src_tgt"][:3, :3]
+    assert torch.allclose(R @ R.T, torch.eye(3), atol=1e-5)
+
+    batch = NeRFDataset.collate_fn([ds[0], ds[1]])
+    assert batch[0]["img"].shape == (2, 3, 32, 40)
+    assert batch[1]["G_src_tgt"].shape == (2, 1, 4, 4)
+
+
+def test_llff_one_train_step(tmp_path):
+    from mine_amd.config import default_config
+    from mine_amd.data.llff import NeRFDataset
+    from mine_amd.engine import SynthesisTask
+
+    _make_scene(str(tmp_path))
+    cfg = default_config(**{
+        "data.name": "llff", "data.img_h": 32, "data.img_w": 40,
+        "mpi.num_bins_coarse": 4, "data.per_gpu_batch_size": 2,
+        "data.visible_point_count": 16,
+        "training.amp_dtype": "fp32",
+        "data.training_set_path": str(tmp_path)})
+    ds = NeRFDataset(cfg, root=str(tmp_path), img_size=(40, 32),
+                     visible_points_count=16)
+    items = NeRFDataset.collate_fn([ds[0], ds[1]])
+    task = SynthesisTask(cfg, device="cpu")
+    loss = task.train_step(items)
+    assert torch.isfinite(loss["loss"])
+
+
+# ---------------------------------------------------------------------------
+# LPIPS
+# ---------------------------------------------------------------------------
+
+
+def test_lpips_metric():
+    from mine_amd.ops.lpips import LPIPS
+    m = LPIPS()
+    assert not m.calibrated
+    a = torch.rand(2, 3, 32, 32, generator=torch.Generator().manual_seed(0))
+    b = torch.rand(2, 3, 32, 32, generator=torch.Generator().manual_seed(1))
+    d_same = m(a, a)
+    d_diff = m(a, b)
+    assert d_same.shape == (2, 1, 1, 1)
+    assert torch.all(d_same.abs() < 1e-6)
+    assert torch.all(d_diff > 1e-4)
+    # deterministic tower
+    m2 = LPIPS()
+    assert torch.allclose(m(a, b), m2(a, b))
+
+
+# ---------------------------------------------------------------------------
+# video CLI pieces
+# ---------------------------------------------------------------------------
+
+
+def test_path_planning_shapes():
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                    "..", "visualizations"))
+    from visualizations.image_to_video import path_planning, synthesize_intrinsics
+    for kind in ("circle", "straight-line", "double-straight-line"):
+        off = path_planning(kind, 0.1, 0.05, 0.2, 24)
+        assert off.shape == (24, 3)
+        assert np.isfinite(off).all()
+    K = synthesize_intrinsics(256, 384)
+    assert K[0, 0] == pytest.approx(192.0)  # 90-degree FoV
+
+
+def test_video_generator_cpu(tmp_path):
+    from mine_amd.config import default_config, RuntimeState
+    from mine_amd.engine import SynthesisTask
+    from visualizations.image_to_video import (VideoGenerator,
+                                               load_source_image,
+                                               path_planning)
+
+    cfg = default_config(**{
+        "data.name": "realestate10k", "data.img_h": 32, "data.img_w": 48,
+        "mpi.num_bins_coarse": 4, "data.per_gpu_batch_size": 1,
+        "data.visible_point_count": 8, "training.amp_dtype": "fp32"})
+    task = SynthesisTask(cfg, state=RuntimeState(), is_val=True, device="cpu")
+    src = load_source_image(None, 32, 48)
+    gen = VideoGenerator(task, cfg, torch.device("cpu"))
+    gen.infer_mpi(src)
+    assert gen.mpi.shape == (1, 4, 32, 48, 4)
+    res = gen.render_pose(np.array([0.02, 0.0, 0.05]))
+    assert res["tgt_imgs_syn"].shape == (1, 3, 32, 48)
+    assert torch.isfinite(res["tgt_imgs_syn"]).all()
+    fps = gen.benchmark_fps(path_planning("circle", 0.05, 0.02, 0.1, 4),
+                            warmup=1)
+    assert fps > 0

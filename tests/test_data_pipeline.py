@@ -231,3 +231,32 @@ def test_video_generator_cpu(tmp_path):
     fps = gen.benchmark_fps(path_planning("circle", 0.05, 0.02, 0.1, 4),
                             warmup=1)
     assert fps > 0
+
+
+# ---------------------------------------------------------------------------
+# COLMAP sqlite helper
+# ---------------------------------------------------------------------------
+
+
+def test_colmap_database(tmp_path):
+    from mine_amd.data.colmap_db import (COLMAPDatabase,
+                                         image_ids_to_pair_id,
+                                         pair_id_to_image_ids)
+    db = COLMAPDatabase.connect(os.path.join(str(tmp_path), "database.db"))
+    db.create_tables()
+    cam = db.add_camera(2, 64, 48, np.array([51.2, 32.0, 24.0, 0.0]))
+    i1 = db.add_image("a.png", cam)
+    i2 = db.add_image("b.png", cam)
+    kp = np.random.default_rng(0).uniform(0, 64, (10, 2)).astype(np.float32)
+    db.add_keypoints(i1, kp)
+    db.add_matches(i1, i2, np.array([[0, 1], [2, 3]], dtype=np.uint32))
+    db.commit()
+
+    rows = db.execute("SELECT rows, cols, data FROM keypoints").fetchone()
+    assert rows[0] == 10 and rows[1] == 2
+    back = np.frombuffer(rows[2], np.float32).reshape(10, 2)
+    assert np.allclose(back, kp)
+    pid = image_ids_to_pair_id(i2, i1)  # order-independent
+    assert pair_id_to_image_ids(pid) == (min(i1, i2), max(i1, i2))
+    assert db.execute("SELECT pair_id FROM matches").fetchone()[0] == pid
+    db.close()

@@ -85,6 +85,14 @@ class SynthesisTask:
         self.device = torch.device(device)
         self.is_gpu = self.device.type == "cuda"
 
+        if self.is_gpu:
+            # MIOpen exhaustive find for the conv stack (the reference's
+            # torch.backends.cudnn.benchmark = True, ref train.py:111-112).
+            # Costs warmup-step time once per shape, then picks the fastest
+            # tuned solver instead of the zero-workspace heuristic fallback.
+            torch.backends.cudnn.benchmark = bool(
+                config.get("training.miopen_benchmark", True))
+
         amp = str(config.get("training.amp_dtype", "bf16")).lower()
         self.amp_dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
                           "fp32": None, "float32": None}[amp]

@@ -182,24 +182,29 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
 
     // ---- pass 2: total cumprod-suffix mass ----
-    // fp64 accumulators: pass 3 computes (TotalP - prefix) / u with
-    // u ~ 1e-6; fp32 cancellation between O(1) terms would amplify to
-    // O(0.1) gradient error after the divide.
+    // fp64 transmittance + accumulators: pass 3 computes
+    // (TotalP - prefix) / u with u ~ 1e-6. The shared-prefix terms of
+    // the two fp64 sums cancel exactly (identical arithmetic in both
+    // passes), so the suffix is accurate to fp64 term rounding; fp32
+    // products would leave O(eps_f32 * |e|) per-term error which the
+    // divide amplifies by 1e6 (0.1-magnitude sigma-grad errors when the
+    // bg-inf depth term makes |e| ~ 1e3).
     double TotalP = 0.0;
-    A = 1.0f;
+    double Ad = 1.0;
     for (int s = 0; s < S; ++s) {
       const float4 px = *reinterpret_cast<const float4*>(
           mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
       const float d = s_depth[s];
       const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
       const float t = __expf(-px.w * delta);
-      const float w = A * (1.0f - t);
+      const float Af = (float)Ad;
+      const float w = Af * (1.0f - t);
       float3 c = make_float3(px.x, px.y, px.z);
       float3 gC = make_float3(0.f, 0.f, 0.f);
       if (BLEND) {
-        c.x = A * I.x + (1.0f - A) * c.x;
-        c.y = A * I.y + (1.0f - A) * c.y;
-        c.z = A * I.z + (1.0f - A) * c.z;
+        c.x = Af * I.x + (1.0f - Af) * c.x;
+        c.y = Af * I.y + (1.0f - Af) * c.y;
+        c.z = Af * I.z + (1.0f - Af) * c.z;
         if (g_blend) {
           const float4 gb = *reinterpret_cast<const float4*>(
               g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
@@ -214,13 +219,13 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
       if (BLEND) {
         dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
       }
-      TotalP += (double)(dA * A);
-      A *= (t + 1e-6f);
+      TotalP += (double)dA * Ad;
+      Ad *= (double)(t + 1e-6f);
     }
 
     // ---- pass 3: emit gradients ----
     double prefix = 0.0;
-    A = 1.0f;
+    Ad = 1.0;
     for (int s = 0; s < S; ++s) {
       const float4 px = *reinterpret_cast<const float4*>(
           mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
@@ -228,14 +233,15 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
       const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
       const float t = __expf(-px.w * delta);
       const float u = t + 1e-6f;
-      const float w = A * (1.0f - t);
+      const float Af = (float)Ad;
+      const float w = Af * (1.0f - t);
       float3 c = make_float3(px.x, px.y, px.z);
       float3 gC = make_float3(0.f, 0.f, 0.f);
       float gCs = 0.0f;
       if (BLEND) {
-        c.x = A * I.x + (1.0f - A) * c.x;
-        c.y = A * I.y + (1.0f - A) * c.y;
-        c.z = A * I.z + (1.0f - A) * c.z;
+        c.x = Af * I.x + (1.0f - Af) * c.x;
+        c.y = Af * I.y + (1.0f - Af) * c.y;
+        c.z = Af * I.z + (1.0f - Af) * c.z;
         if (g_blend) {
           const float4 gb = *reinterpret_cast<const float4*>(
               g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
@@ -251,18 +257,18 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
       if (BLEND) {
         dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
       }
-      prefix += (double)(dA * A);
-      const float dt = -A * e + (float)(TotalP - prefix) / u;
+      prefix += (double)dA * Ad;
+      const float dt = -Af * e + (float)((TotalP - prefix) / (double)u);
       const float dsigma = dt * (-delta * t) + gCs;
       float4 g;
       if (BLEND) {
-        const float oneA = 1.0f - A;
+        const float oneA = 1.0f - Af;
         g = make_float4(oneA * dc.x, oneA * dc.y, oneA * dc.z, dsigma);
       } else {
         g = make_float4(dc.x, dc.y, dc.z, dsigma);
       }
       *reinterpret_cast<float4*>(grad_mpi + mpi_b + ((int64_t)s * HW + pix) * 4) = g;
-      A *= u;
+      Ad *= (double)u;
     }
   }
 }
@@ -491,9 +497,10 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
     const float Wp = Wsum + 1e-5f;
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
 
-    // ---- pass 2 ---- (fp64 suffix accumulators; see src bwd note)
+    // ---- pass 2 ---- (fp64 transmittance + suffix accumulators;
+    // see src bwd note)
     double TotalP = 0.0;
-    A = 1.0f;
+    double Ad = 1.0;
     cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
                        nullptr);
     for (int s = 0; s < S; ++s) {
@@ -511,14 +518,14 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       const float t = __expf(-cur.rgbs.w * delta);
       const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
                       gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
-      TotalP += (double)((1.0f - t) * e * A);
-      A *= (t + 1e-6f);
+      TotalP += (double)((1.0f - t) * e) * Ad;
+      Ad *= (double)(t + 1e-6f);
       cur = nxt;
     }
 
     // ---- pass 3: emit gradients, bilinear scatter ----
     double prefix = 0.0;
-    A = 1.0f;
+    Ad = 1.0;
     TapRef tap;
     cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
                        &tap);
@@ -537,16 +544,17 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       }
       const float t = __expf(-cur.rgbs.w * delta);
       const float u = t + 1e-6f;
-      const float w = A * (1.0f - t);
+      const float Af = (float)Ad;
+      const float w = Af * (1.0f - t);
       const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
                       gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
-      prefix += (double)((1.0f - t) * e * A);
-      const float dt = -A * e + (float)(TotalP - prefix) / u;
+      prefix += (double)((1.0f - t) * e) * Ad;
+      const float dt = -Af * e + (float)((TotalP - prefix) / (double)u);
       // culled sigma contributed nothing -> no gradient through it
       const float dsigma = (cur.v.z < 0.0f) ? 0.0f : dt * (-delta * t);
       scatter4(gm_b + (int64_t)s * HW * 4, tap, W,
                make_float4(w * gR.x, w * gR.y, w * gR.z, dsigma));
-      A *= u;
+      Ad *= (double)u;
       cur = nxt;
       tap = ntap;
     }

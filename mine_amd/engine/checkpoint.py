@@ -24,19 +24,26 @@ def _strip_module_prefix(sd: dict) -> dict:
             for k, v in sd.items()}
 
 
-def save_checkpoint(path: str, backbone, decoder, optimizer=None) -> None:
+def save_checkpoint(path: str, backbone, decoder, optimizer=None,
+                    meta: Optional[dict] = None) -> None:
     state = {"backbone": backbone.state_dict(), "decoder": decoder.state_dict()}
     if optimizer is not None:
         state["optimizer"] = optimizer.state_dict()
+    if meta:
+        # extra key; the reference's loader ignores unknown keys, so the
+        # contract stays compatible. Fixes the reference's resume gap
+        # (epoch/step not persisted, ref synthesis_task.py:661-663).
+        state["meta"] = dict(meta)
     torch.save(state, path)
 
 
 def restore_model(model_path: Optional[str], backbone, decoder, optimizer=None,
-                  logger=None) -> None:
+                  logger=None) -> dict:
+    """Restore; returns the checkpoint's meta dict ({} if absent)."""
     if not model_path:
         if logger:
             logger.info("Not using pre-trained model...")
-        return
+        return {}
     assert os.path.exists(model_path), f"Model {model_path} does not exist!"
     state = torch.load(model_path, map_location="cpu", weights_only=False)
 
@@ -56,3 +63,4 @@ def restore_model(model_path: Optional[str], backbone, decoder, optimizer=None,
             model.load_state_dict(sd, strict=False)
         else:
             model.load_state_dict(sd)
+    return state.get("meta", {})

@@ -118,9 +118,11 @@ class SynthesisTask:
             weight_decay=config["lr.weight_decay"])
 
         # rank-0 restore BEFORE the parameter broadcast (ref CS5)
+        self._restored_meta = {}
         if self.state.is_rank0:
-            restore_model(config.get("training.pretrained_checkpoint_path"),
-                          self.backbone, self.decoder, self.optimizer, logger=logger)
+            self._restored_meta = restore_model(
+                config.get("training.pretrained_checkpoint_path"),
+                self.backbone, self.decoder, self.optimizer, logger=logger) or {}
 
         self.grad_engine = None
         if not is_val:
@@ -439,7 +441,9 @@ class SynthesisTask:
             if step % ckpt_every == 0 and self.state.is_rank0 and \
                     self.state.local_workspace:
                 path = os.path.join(self.state.local_workspace, "checkpoint_latest.pth")
-                save_checkpoint(path, self.backbone, self.decoder, self.optimizer)
+                save_checkpoint(path, self.backbone, self.decoder, self.optimizer,
+                                meta={"epoch": self.current_epoch,
+                                      "global_step": self.global_step})
                 if self.logger:
                     self.logger.info(f"Latest checkpoint saved at {path}")
 
@@ -453,7 +457,15 @@ class SynthesisTask:
                     save_checkpoint(path, self.backbone, self.decoder)
 
     def train(self, train_loader, val_loader=None) -> None:
-        for epoch in range(1, self.config["training.epochs"] + 1):
+        start_epoch = 1
+        if self._restored_meta and bool(self.config.get("training.fine_tune", False)) is False:
+            # resume where the checkpoint left off (improvement over the
+            # reference, which always restarted at epoch 1)
+            start_epoch = int(self._restored_meta.get("epoch", 0)) + 1
+            self.global_step = int(self._restored_meta.get("global_step", 0))
+            for _ in range(1, start_epoch):
+                self.lr_scheduler.step()
+        for epoch in range(start_epoch, self.config["training.epochs"] + 1):
             self.current_epoch = epoch
             self.train_epoch(train_loader, val_loader, epoch)
             self.lr_scheduler.step()

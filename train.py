@@ -68,11 +68,8 @@ def main():
         logger = setup_logger("mine_amd", state.log_file)
         logger.info("Training config: {}".format(dict(config)))
         config.dump_yaml(os.path.join(workspace, "params.yaml"))
-        try:
-            from torch.utils.tensorboard import SummaryWriter
-            state.tb_writer = SummaryWriter(log_dir=workspace)
-        except Exception:
-            logger.info("tensorboard unavailable; scalar logging to file only")
+        from mine_amd.utils.summary import create_summary_writer
+        state.tb_writer = create_summary_writer(workspace)
     state.logger = logger
     if world_size > 1:
         import torch.distributed as dist

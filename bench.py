@@ -34,6 +34,9 @@ def main() -> int:
     p.add_argument("--width", type=int, default=384)
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--dataset", type=str, default="realestate10k")
+    p.add_argument("--timers", action="store_true",
+                   help="print a sync-bracketed per-phase breakdown (diagnostic "
+                        "run only; the extra syncs perturb the headline number)")
     args = p.parse_args()
 
     import torch
@@ -82,6 +85,8 @@ def main() -> int:
     for i in range(args.warmup):
         task.train_step(batches[i % len(batches)])
 
+    if args.timers:
+        task.enable_phase_timers()
     sync()
     t0 = time.perf_counter()
     for i in range(args.steps):
@@ -97,6 +102,10 @@ def main() -> int:
         elapsed = float(t.item())
 
     imgs_per_sec = world_size * args.batch * args.steps / elapsed
+    if args.timers and rank == 0:
+        phases = {k: round(v * 1000.0, 1)
+                  for k, v in task.pop_phase_times().items()}
+        print(json.dumps({"phase_ms": phases}), file=sys.stderr)
     if rank == 0:
         result = {
             "metric": "train imgs/sec (whole node) RealEstate10K 384x256 N=64",

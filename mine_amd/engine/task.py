@@ -406,8 +406,19 @@ class SynthesisTask:
     # ------------------------------------------------------------------
     def train_step(self, items) -> dict:
         """One optimization step; returns the loss dict."""
+        timers = getattr(self, "phase_timers", None)
+
+        def mark(name):
+            if timers is not None:
+                if self.is_gpu:
+                    torch.cuda.synchronize()
+                timers.setdefault(name, []).append(time.perf_counter())
+
+        mark("t0")
         self.set_data(items)
+        mark("set_data")
         loss_dict, _ = self.loss_fcn(is_val=False)
+        mark("forward")
         if self.grad_engine is not None:
             self.grad_engine.zero_grad()
         else:
@@ -415,8 +426,31 @@ class SynthesisTask:
         loss_dict["loss"].backward()
         if self.grad_engine is not None:
             self.grad_engine.finish_step()
+        mark("backward")
         self.optimizer.step()
+        mark("optimizer")
         return loss_dict
+
+    def enable_phase_timers(self) -> None:
+        """Sync-bracketed per-phase wall times; read with pop_phase_times()."""
+        self.phase_timers = {}
+
+    def pop_phase_times(self) -> dict:
+        """Mean seconds per phase since the last call."""
+        t = getattr(self, "phase_timers", None)
+        if not t or "t0" not in t:
+            return {}
+        names = ["set_data", "forward", "backward", "optimizer"]
+        out = {}
+        prev = t["t0"]
+        for n in names:
+            cur = t.get(n)
+            if cur is None or len(cur) != len(prev):
+                break
+            out[n] = sum(b - a for a, b in zip(prev, cur)) / len(cur)
+            prev = cur
+        self.phase_timers = {}
+        return out
 
     def train_epoch(self, train_loader, val_loader, epoch: int) -> None:
         cfg = self.config

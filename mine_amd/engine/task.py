@@ -86,12 +86,14 @@ class SynthesisTask:
         self.is_gpu = self.device.type == "cuda"
 
         if self.is_gpu:
-            # MIOpen exhaustive find for the conv stack (the reference's
-            # torch.backends.cudnn.benchmark = True, ref train.py:111-112).
-            # Costs warmup-step time once per shape, then picks the fastest
-            # tuned solver instead of the zero-workspace heuristic fallback.
+            # The reference sets torch.backends.cudnn.benchmark = True
+            # (ref train.py:111-112). On this ROCm 7.2 / MIOpen build the
+            # exhaustive-find path it triggers memory-faults the GPU
+            # (observed on MI355X during find at the flagship shapes), so
+            # it is OFF by default; MIOPEN_FIND_MODE (set to FAST in
+            # mine_amd/__init__.py) governs solver selection instead.
             torch.backends.cudnn.benchmark = bool(
-                config.get("training.miopen_benchmark", True))
+                config.get("training.miopen_benchmark", False))
 
         amp = str(config.get("training.amp_dtype", "bf16")).lower()
         self.amp_dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,

@@ -119,7 +119,7 @@ def src_plane_xyz(meshgrid: torch.Tensor, disparity: torch.Tensor,
     B, S = disparity.shape
     H, W = meshgrid.shape[-2:]
     depth = torch.reciprocal(disparity)  # BxS
-    rays = torch.matmul(K_inv, meshgrid.reshape(3, -1))  # Bx3xHW
+    rays = torch.matmul(K_inv, meshgrid.reshape(3, -1).to(K_inv.dtype))  # Bx3xHW
     xyz = rays.unsqueeze(1) * depth[:, :, None, None]  # BxSx3xHW
     return xyz.reshape(B, S, 3, H, W)
 
@@ -226,8 +226,8 @@ def homography_grid_sample(src: torch.Tensor, H_src_tgt: torch.Tensor,
     Returns (warped NxCxH_outxW_out, valid mask NxH_outxW_out bool).
     """
     N, C, H_src, W_src = src.shape
-    grid = make_meshgrid(H_out, W_out, device=src.device)  # 3xHxW
-    mapped = torch.matmul(H_src_tgt, grid.reshape(3, -1))  # Nx3xHW
+    grid = make_meshgrid(H_out, W_out, device=src.device).to(src.dtype)  # 3xHxW
+    mapped = torch.matmul(H_src_tgt.to(src.dtype), grid.reshape(3, -1))  # Nx3xHW
     mapped = mapped.reshape(N, 3, H_out, W_out).permute(0, 2, 3, 1)  # NxHxWx3
     uv = mapped[..., :2] / mapped[..., 2:]
 

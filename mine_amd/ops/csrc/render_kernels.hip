@@ -123,6 +123,48 @@ src_composite_fwd_kernel(const float* __restrict__ mpi,
   }
 }
 
+// Per-plane backward term, shared by passes 2 and 3.
+//
+// noinline is LOAD-BEARING: the two passes telescope fp64 prefix sums of
+// dA*A against each other and divide the residue by u ~ 1e-6. If the
+// passes inline this computation separately, FMA contraction may fuse
+// differently in the two contexts, the terms stop being bitwise equal,
+// and the nonzero residue at late planes amplifies into O(0.1) sigma
+// gradients. One out-of-line instance -> identical bits -> exact
+// cancellation.
+struct PlaneTerm {
+  float t;      // transparency exp(-sigma*delta)
+  float e;      // dL/dw
+  float dA;     // dL/dA_s (direct)
+  float dcx, dcy, dcz;  // dL/dc_s
+};
+
+template <bool BLEND, bool BG_INF>
+__device__ __attribute__((noinline)) PlaneTerm
+src_plane_term(float4 px, float Af, float3 I, float3 gR, float gD,
+               float d, float delta, float D, float Wp, float3 gC) {
+  PlaneTerm r;
+  r.t = __expf(-px.w * delta);
+  const float w = Af * (1.0f - r.t);
+  float3 c = make_float3(px.x, px.y, px.z);
+  if (BLEND) {
+    c.x = Af * I.x + (1.0f - Af) * c.x;
+    c.y = Af * I.y + (1.0f - Af) * c.y;
+    c.z = Af * I.z + (1.0f - Af) * c.z;
+  }
+  r.dcx = w * gR.x + gC.x;
+  r.dcy = w * gR.y + gC.y;
+  r.dcz = w * gR.z + gC.z;
+  r.e = c.x * gR.x + c.y * gR.y + c.z * gR.z +
+        gD * (BG_INF ? (d - 1000.0f) : (d - D) / Wp);
+  float dA = (1.0f - r.t) * r.e;
+  if (BLEND) {
+    dA += (I.x - px.x) * r.dcx + (I.y - px.y) * r.dcy + (I.z - px.z) * r.dcz;
+  }
+  r.dA = dA;
+  return r;
+}
+
 template <bool BLEND, bool BG_INF>
 __global__ void __launch_bounds__(kBlock)
 src_composite_bwd_kernel(const float* __restrict__ mpi,
@@ -184,11 +226,12 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
     // ---- pass 2: total cumprod-suffix mass ----
     // fp64 transmittance + accumulators: pass 3 computes
     // (TotalP - prefix) / u with u ~ 1e-6. The shared-prefix terms of
-    // the two fp64 sums cancel exactly (identical arithmetic in both
-    // passes), so the suffix is accurate to fp64 term rounding; fp32
-    // products would leave O(eps_f32 * |e|) per-term error which the
-    // divide amplifies by 1e6 (0.1-magnitude sigma-grad errors when the
-    // bg-inf depth term makes |e| ~ 1e3).
+    // the two fp64 sums must cancel EXACTLY, so both passes obtain the
+    // per-plane term from the single noinline src_plane_term instance
+    // (bitwise-identical values); fp32 term differences would leave
+    // O(eps_f32 * |e|) residues which the divide amplifies by up to 1e6
+    // (0.1-magnitude sigma-grad errors when the bg-inf depth term makes
+    // |e| ~ 1e3).
     double TotalP = 0.0;
     double Ad = 1.0;
     for (int s = 0; s < S; ++s) {
@@ -196,31 +239,16 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
           mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
       const float d = s_depth[s];
       const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
-      const float t = __expf(-px.w * delta);
-      const float Af = (float)Ad;
-      const float w = Af * (1.0f - t);
-      float3 c = make_float3(px.x, px.y, px.z);
       float3 gC = make_float3(0.f, 0.f, 0.f);
-      if (BLEND) {
-        c.x = Af * I.x + (1.0f - Af) * c.x;
-        c.y = Af * I.y + (1.0f - Af) * c.y;
-        c.z = Af * I.z + (1.0f - Af) * c.z;
-        if (g_blend) {
-          const float4 gb = *reinterpret_cast<const float4*>(
-              g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
-          gC = make_float3(gb.x, gb.y, gb.z);
-        }
+      if (BLEND && g_blend) {
+        const float4 gb = *reinterpret_cast<const float4*>(
+            g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
+        gC = make_float3(gb.x, gb.y, gb.z);
       }
-      const float3 dc = make_float3(w * gR.x + gC.x, w * gR.y + gC.y,
-                                    w * gR.z + gC.z);
-      const float e = c.x * gR.x + c.y * gR.y + c.z * gR.z +
-                      gD * (BG_INF ? (d - 1000.0f) : (d - D) / Wp);
-      float dA = (1.0f - t) * e;
-      if (BLEND) {
-        dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
-      }
-      TotalP += (double)dA * Ad;
-      Ad *= (double)(t + 1e-6f);
+      const PlaneTerm pt = src_plane_term<BLEND, BG_INF>(
+          px, (float)Ad, I, gR, gD, d, delta, D, Wp, gC);
+      TotalP += (double)pt.dA * Ad;
+      Ad *= (double)(pt.t + 1e-6f);
     }
 
     // ---- pass 3: emit gradients ----
@@ -231,41 +259,27 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
           mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
       const float d = s_depth[s];
       const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
-      const float t = __expf(-px.w * delta);
-      const float u = t + 1e-6f;
-      const float Af = (float)Ad;
-      const float w = Af * (1.0f - t);
-      float3 c = make_float3(px.x, px.y, px.z);
       float3 gC = make_float3(0.f, 0.f, 0.f);
       float gCs = 0.0f;
-      if (BLEND) {
-        c.x = Af * I.x + (1.0f - Af) * c.x;
-        c.y = Af * I.y + (1.0f - Af) * c.y;
-        c.z = Af * I.z + (1.0f - Af) * c.z;
-        if (g_blend) {
-          const float4 gb = *reinterpret_cast<const float4*>(
-              g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
-          gC = make_float3(gb.x, gb.y, gb.z);
-          gCs = gb.w;
-        }
+      if (BLEND && g_blend) {
+        const float4 gb = *reinterpret_cast<const float4*>(
+            g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
+        gC = make_float3(gb.x, gb.y, gb.z);
+        gCs = gb.w;
       }
-      const float3 dc = make_float3(w * gR.x + gC.x, w * gR.y + gC.y,
-                                    w * gR.z + gC.z);
-      const float e = c.x * gR.x + c.y * gR.y + c.z * gR.z +
-                      gD * (BG_INF ? (d - 1000.0f) : (d - D) / Wp);
-      float dA = (1.0f - t) * e;
-      if (BLEND) {
-        dA += (I.x - px.x) * dc.x + (I.y - px.y) * dc.y + (I.z - px.z) * dc.z;
-      }
-      prefix += (double)dA * Ad;
-      const float dt = -Af * e + (float)((TotalP - prefix) / (double)u);
-      const float dsigma = dt * (-delta * t) + gCs;
+      const float Af = (float)Ad;
+      const PlaneTerm pt = src_plane_term<BLEND, BG_INF>(
+          px, Af, I, gR, gD, d, delta, D, Wp, gC);
+      const float u = pt.t + 1e-6f;
+      prefix += (double)pt.dA * Ad;
+      const float dt = -Af * pt.e + (float)((TotalP - prefix) / (double)u);
+      const float dsigma = dt * (-delta * pt.t) + gCs;
       float4 g;
       if (BLEND) {
         const float oneA = 1.0f - Af;
-        g = make_float4(oneA * dc.x, oneA * dc.y, oneA * dc.z, dsigma);
+        g = make_float4(oneA * pt.dcx, oneA * pt.dcy, oneA * pt.dcz, dsigma);
       } else {
-        g = make_float4(dc.x, dc.y, dc.z, dsigma);
+        g = make_float4(pt.dcx, pt.dcy, pt.dcz, dsigma);
       }
       *reinterpret_cast<float4*>(grad_mpi + mpi_b + ((int64_t)s * HW + pix) * 4) = g;
       Ad *= (double)u;
@@ -432,6 +446,18 @@ DEV void scatter4(float* base, const TapRef& t, int W, float4 g) {
 #undef SC
 }
 
+// Per-plane (t, e) for the tgt backward — noinline for the same exact-
+// cancellation reason as src_plane_term above.
+template <bool BG_INF>
+__device__ __attribute__((noinline)) float2
+tgt_plane_term(float4 rgbs, float vz, float delta, float3 gR, float gD,
+               float D, float Wp) {
+  const float t = __expf(-rgbs.w * delta);
+  const float e = rgbs.x * gR.x + rgbs.y * gR.y + rgbs.z * gR.z +
+                  gD * (BG_INF ? (vz - 1000.0f) : (vz - D) / Wp);
+  return make_float2(t, e);
+}
+
 template <bool BG_INF>
 __global__ void __launch_bounds__(kBlock)
 tgt_composite_bwd_kernel(const float* __restrict__ mpi,
@@ -515,11 +541,10 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       } else {
         delta = 1e3f;
       }
-      const float t = __expf(-cur.rgbs.w * delta);
-      const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
-                      gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
-      TotalP += (double)((1.0f - t) * e) * Ad;
-      Ad *= (double)(t + 1e-6f);
+      const float2 te = tgt_plane_term<BG_INF>(cur.rgbs, cur.v.z, delta,
+                                               gR, gD, D, Wp);
+      TotalP += (double)((1.0f - te.x) * te.y) * Ad;
+      Ad *= (double)(te.x + 1e-6f);
       cur = nxt;
     }
 
@@ -542,14 +567,14 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       } else {
         delta = 1e3f;
       }
-      const float t = __expf(-cur.rgbs.w * delta);
+      const float2 te = tgt_plane_term<BG_INF>(cur.rgbs, cur.v.z, delta,
+                                               gR, gD, D, Wp);
+      const float t = te.x;
       const float u = t + 1e-6f;
       const float Af = (float)Ad;
       const float w = Af * (1.0f - t);
-      const float e = cur.rgbs.x * gR.x + cur.rgbs.y * gR.y + cur.rgbs.z * gR.z +
-                      gD * (BG_INF ? (cur.v.z - 1000.0f) : (cur.v.z - D) / Wp);
-      prefix += (double)((1.0f - t) * e) * Ad;
-      const float dt = -Af * e + (float)((TotalP - prefix) / (double)u);
+      prefix += (double)((1.0f - t) * te.y) * Ad;
+      const float dt = -Af * te.y + (float)((TotalP - prefix) / (double)u);
       // culled sigma contributed nothing -> no gradient through it
       const float dsigma = (cur.v.z < 0.0f) ? 0.0f : dt * (-delta * t);
       scatter4(gm_b + (int64_t)s * HW * 4, tap, W,

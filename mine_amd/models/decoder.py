@@ -29,11 +29,15 @@ from mine_amd.utils.embedder import PositionalEncoder
 
 
 class ConvBlock(nn.Module):
-    """ReflectionPad(1) + 3x3 conv + BN + ELU (ref monodepth2/layers.py:106-138)."""
+    """ReflectionPad(1) + 3x3 conv + BN + ELU (ref monodepth2/layers.py:106-138).
+
+    The pad runs on the HIP gather kernels (mine_amd/ops/pad.py) — the
+    eager pad pair was ~24% of the train step (profiles/r01_*)."""
 
     def __init__(self, in_ch: int, out_ch: int):
         super().__init__()
-        self.pad = nn.ReflectionPad2d(1)
+        from mine_amd.ops.pad import ReflectionPad2d
+        self.pad = ReflectionPad2d(1)
         self.conv = nn.Conv2d(in_ch, out_ch, 3)
         self.bn = Fp32BatchNorm2d(out_ch)
         self.act = nn.ELU(inplace=True)
@@ -55,7 +59,8 @@ class Conv3x3Refl(nn.Module):
 
     def __init__(self, in_ch: int, out_ch: int):
         super().__init__()
-        self.pad = nn.ReflectionPad2d(1)
+        from mine_amd.ops.pad import ReflectionPad2d
+        self.pad = ReflectionPad2d(1)
         self.conv = nn.Conv2d(in_ch, out_ch, 3)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

@@ -25,6 +25,14 @@ void mine_tgt_composite_bwd(const float*, const float*, const float*,
                             const float*, const float*, const float*,
                             const float*, float*, int, int, int, int, int,
                             hipStream_t);
+void mine_reflect_pad_fwd_f32(const float*, float*, int, int, int, int, int,
+                              hipStream_t);
+void mine_reflect_pad_fwd_bf16(const void*, void*, int, int, int, int, int,
+                               hipStream_t);
+void mine_reflect_pad_bwd_f32(const float*, float*, int, int, int, int, int,
+                              hipStream_t);
+void mine_reflect_pad_bwd_bf16(const void*, void*, int, int, int, int, int,
+                               hipStream_t);
 void mine_ssim_set_window(const float*);
 void mine_ssim_fwd(const float*, const float*, float*, int, int, int,
                    hipStream_t);
@@ -120,6 +128,45 @@ at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor m,
 }
 
 // --------------------------------------------------------------------------
+// reflection pad — logical (N,H,W,C) layout (see pad_kernels.hip).
+// The caller (mine_amd/ops/pad.py) maps channels_last / NCHW onto it.
+
+at::Tensor reflect_pad_fwd(at::Tensor in, int64_t N, int64_t H, int64_t W,
+                           int64_t C, int64_t pad) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous());
+  TORCH_CHECK(in.numel() == N * H * W * C, "bad logical shape");
+  TORCH_CHECK(H > pad && W > pad, "pad must be < spatial size");
+  auto out = at::empty({N * (H + 2 * pad) * (W + 2 * pad) * C}, in.options());
+  if (in.scalar_type() == at::kFloat) {
+    mine_reflect_pad_fwd_f32(in.data_ptr<float>(), out.data_ptr<float>(),
+                             N, H, W, C, pad, stream());
+  } else if (in.scalar_type() == at::kBFloat16) {
+    mine_reflect_pad_fwd_bf16(in.data_ptr(), out.data_ptr(),
+                              N, H, W, C, pad, stream());
+  } else {
+    TORCH_CHECK(false, "reflect_pad: dtype must be f32 or bf16");
+  }
+  return out;
+}
+
+at::Tensor reflect_pad_bwd(at::Tensor gout, int64_t N, int64_t H, int64_t W,
+                           int64_t C, int64_t pad) {
+  TORCH_CHECK(gout.is_cuda() && gout.is_contiguous());
+  TORCH_CHECK(gout.numel() == N * (H + 2 * pad) * (W + 2 * pad) * C);
+  auto gin = at::empty({N * H * W * C}, gout.options());
+  if (gout.scalar_type() == at::kFloat) {
+    mine_reflect_pad_bwd_f32(gout.data_ptr<float>(), gin.data_ptr<float>(),
+                             N, H, W, C, pad, stream());
+  } else if (gout.scalar_type() == at::kBFloat16) {
+    mine_reflect_pad_bwd_bf16(gout.data_ptr(), gin.data_ptr(),
+                              N, H, W, C, pad, stream());
+  } else {
+    TORCH_CHECK(false, "reflect_pad: dtype must be f32 or bf16");
+  }
+  return gin;
+}
+
+// --------------------------------------------------------------------------
 
 bool g_window_set = false;
 
@@ -178,4 +225,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("tgt_composite_bwd", &tgt_composite_bwd);
   mod.def("ssim_fwd", &ssim_fwd);
   mod.def("ssim_bwd", &ssim_bwd);
+  mod.def("reflect_pad_fwd", &reflect_pad_fwd,
+          "gather reflection pad over logical (N,H,W,C)");
+  mod.def("reflect_pad_bwd", &reflect_pad_bwd,
+          "gather (atomic-free) reflection pad backward");
 }

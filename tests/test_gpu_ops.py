@@ -224,3 +224,43 @@ def test_native_extension_is_loaded_on_gpu():
     assert has_extension()
     ext = get_extension()
     assert "mine_amd" in ext.__file__
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("channels_last", [False, True])
+def test_reflection_pad_matches_torch(dtype, channels_last):
+    from mine_amd.ops.pad import reflection_pad2d
+    import torch.nn.functional as F
+
+    g = torch.Generator().manual_seed(3)
+    x0 = torch.randn(3, 7, 10, 14, generator=g)
+    x = x0.to("cuda:0", dtype)
+    if channels_last:
+        x = x.contiguous(memory_format=torch.channels_last)
+    x = x.requires_grad_(True)
+    y = reflection_pad2d(x, 1)
+    w = torch.randn(3, 7, 12, 16, generator=g).to("cuda:0", dtype)
+    (y * w).sum().backward()
+
+    xr = x0.clone().to(dtype).requires_grad_(True)
+    yr = F.pad(xr, (1, 1, 1, 1), mode="reflect")
+    (yr * w.cpu()).sum().backward()
+
+    torch.testing.assert_close(y.float().cpu(), yr.float(), rtol=0, atol=0)
+    tol = 1e-5 if dtype == torch.float32 else 5e-2
+    torch.testing.assert_close(x.grad.float().cpu(), xr.grad.float(),
+                               rtol=tol, atol=tol)
+
+
+def test_reflection_pad_pad2():
+    from mine_amd.ops.pad import reflection_pad2d
+    import torch.nn.functional as F
+    g = torch.Generator().manual_seed(4)
+    x0 = torch.randn(2, 3, 6, 9, generator=g)
+    x = x0.to("cuda:0").requires_grad_(True)
+    y = reflection_pad2d(x, 2)
+    y.sum().backward()
+    xr = x0.clone().requires_grad_(True)
+    F.pad(xr, (2, 2, 2, 2), mode="reflect").sum().backward()
+    torch.testing.assert_close(y.cpu(), F.pad(x0, (2, 2, 2, 2), mode="reflect"))
+    torch.testing.assert_close(x.grad.cpu(), xr.grad, rtol=1e-6, atol=1e-6)

@@ -22,7 +22,6 @@ from typing import Dict, List, Sequence, Tuple
 import torch
 import torch.nn as nn
 
-from mine_amd.models.norm import Fp32BatchNorm2d
 import torch.nn.functional as F
 
 from mine_amd.utils.embedder import PositionalEncoder
@@ -36,21 +35,21 @@ class ConvBlock(nn.Module):
 
     def __init__(self, in_ch: int, out_ch: int):
         super().__init__()
+        from mine_amd.ops.bn import FusedBNAct
         from mine_amd.ops.pad import ReflectionPad2d
         self.pad = ReflectionPad2d(1)
         self.conv = nn.Conv2d(in_ch, out_ch, 3)
-        self.bn = Fp32BatchNorm2d(out_ch)
-        self.act = nn.ELU(inplace=True)
+        self.bn = FusedBNAct(out_ch, act="elu")
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.act(self.bn(self.conv(self.pad(x))))
+        return self.bn(self.conv(self.pad(x)))
 
 
 def _neck_conv(in_ch: int, out_ch: int, k: int) -> nn.Sequential:
+    from mine_amd.ops.bn import FusedBNAct
     return nn.Sequential(
         nn.Conv2d(in_ch, out_ch, k, stride=1, padding=(k - 1) // 2, bias=False),
-        Fp32BatchNorm2d(out_ch),
-        nn.LeakyReLU(0.1, inplace=True),
+        FusedBNAct(out_ch, act="lrelu"),
     )
 
 

@@ -24,7 +24,7 @@ from typing import List
 import torch
 import torch.nn as nn
 
-from mine_amd.models.norm import Fp32BatchNorm2d
+from mine_amd.ops.bn import FusedBNAct
 
 _IMAGENET_MEAN = (0.485, 0.456, 0.406)
 _IMAGENET_STD = (0.229, 0.224, 0.225)
@@ -38,20 +38,19 @@ class Bottleneck(nn.Module):
         super().__init__()
         out_ch = planes * self.expansion
         self.conv1 = nn.Conv2d(in_ch, planes, 1, bias=False)
-        self.bn1 = Fp32BatchNorm2d(planes)
+        self.bn1 = FusedBNAct(planes, act="relu")
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
-        self.bn2 = Fp32BatchNorm2d(planes)
+        self.bn2 = FusedBNAct(planes, act="relu")
         self.conv3 = nn.Conv2d(planes, out_ch, 1, bias=False)
-        self.bn3 = Fp32BatchNorm2d(out_ch)
-        self.relu = nn.ReLU(inplace=True)
+        # residual join fused into the bn3 epilogue: relu(bn3(conv3) + id)
+        self.bn3 = FusedBNAct(out_ch, act="add_relu")
         self.downsample = downsample
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        return self.relu(out + identity)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), identity)
 
 
 class ResNetEncoder(nn.Module):
@@ -66,7 +65,7 @@ class ResNetEncoder(nn.Module):
         self.num_ch_enc = [64, 256, 512, 1024, 2048]
 
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = Fp32BatchNorm2d(64)
+        self.bn1 = FusedBNAct(64, act="relu")
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
 
@@ -98,7 +97,7 @@ class ResNetEncoder(nn.Module):
         if stride != 1 or self.in_ch != out_ch:
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
-                Fp32BatchNorm2d(out_ch),
+                FusedBNAct(out_ch, act="none"),
             )
         layers = [Bottleneck(self.in_ch, planes, stride, downsample)]
         self.in_ch = out_ch
@@ -109,8 +108,7 @@ class ResNetEncoder(nn.Module):
     def forward(self, img: torch.Tensor) -> List[torch.Tensor]:
         """img: Bx3xHxW in [0,1]. Returns 5 taps at strides 2..32."""
         x = (img - self.img_mean) / self.img_std
-        x = self.relu(self.bn1(self.conv1(x)))
-        conv1_out = x
+        conv1_out = self.bn1(self.conv1(x))  # relu fused in bn1
         b1 = self.layer1(self.maxpool(conv1_out))
         b2 = self.layer2(b1)
         b3 = self.layer3(b2)

@@ -33,6 +33,32 @@ void mine_reflect_pad_bwd_f32(const float*, float*, int, int, int, int, int,
                               hipStream_t);
 void mine_reflect_pad_bwd_bf16(const void*, void*, int, int, int, int, int,
                                hipStream_t);
+void mine_bn_stats_f32(const void*, float*, int64_t, int, hipStream_t);
+void mine_bn_stats_bf16(const void*, float*, int64_t, int, hipStream_t);
+void mine_bn_finalize(const float*, float*, float*, float*, float*, int64_t,
+                      int, float, float, hipStream_t);
+void mine_bn_act_fwd_f32(const void*, const void*, const float*, const float*,
+                         const float*, const float*, void*, int64_t, int, int,
+                         hipStream_t);
+void mine_bn_act_fwd_bf16(const void*, const void*, const float*, const float*,
+                          const float*, const float*, void*, int64_t, int, int,
+                          hipStream_t);
+void mine_bn_act_bwd_reduce_f32(const void*, const void*, const void*,
+                                const float*, const float*, const float*,
+                                const float*, float*, int64_t, int, int,
+                                hipStream_t);
+void mine_bn_act_bwd_reduce_bf16(const void*, const void*, const void*,
+                                 const float*, const float*, const float*,
+                                 const float*, float*, int64_t, int, int,
+                                 hipStream_t);
+void mine_bn_act_bwd_dx_f32(const void*, const void*, const void*,
+                            const float*, const float*, const float*,
+                            const float*, const float*, void*, void*, int64_t,
+                            int, int, hipStream_t);
+void mine_bn_act_bwd_dx_bf16(const void*, const void*, const void*,
+                             const float*, const float*, const float*,
+                             const float*, const float*, void*, void*, int64_t,
+                             int, int, hipStream_t);
 void mine_ssim_set_window(const float*);
 void mine_ssim_fwd(const float*, const float*, float*, int, int, int,
                    hipStream_t);
@@ -167,6 +193,77 @@ at::Tensor reflect_pad_bwd(at::Tensor gout, int64_t N, int64_t H, int64_t W,
 }
 
 // --------------------------------------------------------------------------
+// fused BatchNorm + activation (see bn_kernels.hip). Tensors arrive as
+// flat logical (M, C) channels_last views; f32 or bf16.
+
+#define BN_DISPATCH(FN, T, ...)                                   \
+  do {                                                            \
+    if ((T) == at::kFloat) FN##_f32(__VA_ARGS__);                 \
+    else if ((T) == at::kBFloat16) FN##_bf16(__VA_ARGS__);        \
+    else TORCH_CHECK(false, "bn: dtype must be f32 or bf16");     \
+  } while (0)
+
+std::vector<at::Tensor> bn_stats(at::Tensor x, int64_t M, int64_t C,
+                                 at::Tensor running_mean,
+                                 at::Tensor running_var, double eps,
+                                 double momentum) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.numel() == M * C);
+  auto f32 = x.options().dtype(at::kFloat);
+  auto sums = at::zeros({2 * C}, f32);
+  BN_DISPATCH(mine_bn_stats, x.scalar_type(), x.data_ptr(),
+              sums.data_ptr<float>(), M, (int)C, stream());
+  auto mean = at::empty({C}, f32);
+  auto invstd = at::empty({C}, f32);
+  const bool track = running_mean.numel() > 0;
+  mine_bn_finalize(sums.data_ptr<float>(), mean.data_ptr<float>(),
+                   invstd.data_ptr<float>(),
+                   track ? running_mean.data_ptr<float>() : nullptr,
+                   track ? running_var.data_ptr<float>() : nullptr,
+                   M, (int)C, (float)eps, (float)momentum, stream());
+  return {mean, invstd};
+}
+
+at::Tensor bn_act_fwd(at::Tensor x, at::Tensor res, at::Tensor mean,
+                      at::Tensor invstd, at::Tensor gamma, at::Tensor beta,
+                      int64_t M, int64_t C, int64_t act) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.numel() == M * C);
+  auto y = at::empty_like(x);
+  BN_DISPATCH(mine_bn_act_fwd, x.scalar_type(), x.data_ptr(),
+              res.numel() ? res.data_ptr() : nullptr,
+              mean.data_ptr<float>(), invstd.data_ptr<float>(),
+              gamma.data_ptr<float>(), beta.data_ptr<float>(), y.data_ptr(),
+              M, (int)C, (int)act, stream());
+  return y;
+}
+
+std::vector<at::Tensor> bn_act_bwd(at::Tensor x, at::Tensor res,
+                                   at::Tensor gy, at::Tensor mean,
+                                   at::Tensor invstd, at::Tensor gamma,
+                                   at::Tensor beta, int64_t M, int64_t C,
+                                   int64_t act) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && gy.is_contiguous());
+  auto f32 = x.options().dtype(at::kFloat);
+  auto red = at::zeros({2 * C}, f32);  // (dbeta, dgamma)
+  const bool add_relu = act == 4;
+  BN_DISPATCH(mine_bn_act_bwd_reduce, x.scalar_type(), x.data_ptr(),
+              add_relu ? res.data_ptr() : nullptr, gy.data_ptr(),
+              mean.data_ptr<float>(), invstd.data_ptr<float>(),
+              gamma.data_ptr<float>(), beta.data_ptr<float>(),
+              red.data_ptr<float>(), M, (int)C, (int)act, stream());
+  auto dx = at::empty_like(x);
+  auto dres = add_relu ? at::empty_like(x) : at::empty({0}, x.options());
+  BN_DISPATCH(mine_bn_act_bwd_dx, x.scalar_type(), x.data_ptr(),
+              add_relu ? res.data_ptr() : nullptr, gy.data_ptr(),
+              mean.data_ptr<float>(), invstd.data_ptr<float>(),
+              gamma.data_ptr<float>(), beta.data_ptr<float>(),
+              red.data_ptr<float>(), dx.data_ptr(),
+              add_relu ? dres.data_ptr() : nullptr, M, (int)C, (int)act,
+              stream());
+  // dbeta = red[:C], dgamma = red[C:]
+  return {dx, dres, red.narrow(0, C, C).clone(), red.narrow(0, 0, C).clone()};
+}
+
+// --------------------------------------------------------------------------
 
 bool g_window_set = false;
 
@@ -229,4 +326,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "gather reflection pad over logical (N,H,W,C)");
   mod.def("reflect_pad_bwd", &reflect_pad_bwd,
           "gather (atomic-free) reflection pad backward");
+  mod.def("bn_stats", &bn_stats,
+          "per-channel mean/invstd + running-stat update");
+  mod.def("bn_act_fwd", &bn_act_fwd, "fused normalize + activation");
+  mod.def("bn_act_bwd", &bn_act_bwd,
+          "fused BN+act backward -> dx, dres, dgamma, dbeta");
 }

@@ -264,3 +264,101 @@ def test_reflection_pad_pad2():
     F.pad(xr, (2, 2, 2, 2), mode="reflect").sum().backward()
     torch.testing.assert_close(y.cpu(), F.pad(x0, (2, 2, 2, 2), mode="reflect"))
     torch.testing.assert_close(x.grad.cpu(), xr.grad, rtol=1e-6, atol=1e-6)
+
+
+@pytest.mark.parametrize("act", ["none", "relu", "lrelu", "elu"])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_bn_act_matches_torch(act, dtype):
+    from mine_amd.ops.bn import FusedBNAct, _act_eager
+    import torch.nn.functional as F
+
+    g = torch.Generator().manual_seed(11)
+    B, C, H, W = 4, 12, 17, 23
+    x0 = torch.randn(B, C, H, W, generator=g)
+    w0 = torch.rand(C, generator=g) + 0.5
+    b0 = torch.randn(C, generator=g) * 0.2
+    gy = torch.randn(B, C, H, W, generator=g)
+
+    # fused path (GPU, channels_last)
+    m = FusedBNAct(C, act=act).to("cuda:0").train()
+    with torch.no_grad():
+        m.weight.copy_(w0)
+        m.bias.copy_(b0)
+    x = x0.to("cuda:0", dtype).contiguous(memory_format=torch.channels_last
+                                          ).requires_grad_(True)
+    y = m(x)
+    (y.float() * gy.to("cuda:0")).sum().backward()
+
+    # fp32 eager reference (CPU)
+    xr = x0.clone().requires_grad_(True)
+    wr = w0.clone().requires_grad_(True)
+    br = b0.clone().requires_grad_(True)
+    rm = torch.zeros(C)
+    rv = torch.ones(C)
+    zr = F.batch_norm(xr, rm, rv, wr, br, True, 0.1, 1e-5)
+    yr = _act_eager(act, zr)
+    (yr * gy).sum().backward()
+
+    tol = dict(rtol=1e-4, atol=1e-4) if dtype == torch.float32 else \
+        dict(rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(y.float().cpu(), yr, **tol)
+    torch.testing.assert_close(x.grad.float().cpu(), xr.grad, **tol)
+    torch.testing.assert_close(m.weight.grad.cpu(), wr.grad,
+                               rtol=1e-2 if dtype != torch.float32 else 1e-4,
+                               atol=1e-2 if dtype != torch.float32 else 1e-3)
+    torch.testing.assert_close(m.bias.grad.cpu(), br.grad,
+                               rtol=1e-2 if dtype != torch.float32 else 1e-4,
+                               atol=1e-2 if dtype != torch.float32 else 1e-3)
+    torch.testing.assert_close(m.running_mean.cpu(), rm, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(m.running_var.cpu(), rv, rtol=1e-3, atol=1e-3)
+
+
+def test_fused_bn_add_relu_matches_torch():
+    from mine_amd.ops.bn import FusedBNAct
+    import torch.nn.functional as F
+
+    g = torch.Generator().manual_seed(12)
+    B, C, H, W = 3, 8, 14, 19
+    x0 = torch.randn(B, C, H, W, generator=g)
+    r0 = torch.randn(B, C, H, W, generator=g)
+    gy = torch.randn(B, C, H, W, generator=g)
+
+    m = FusedBNAct(C, act="add_relu").to("cuda:0").train()
+    x = x0.to("cuda:0").contiguous(memory_format=torch.channels_last
+                                   ).requires_grad_(True)
+    r = r0.to("cuda:0").contiguous(memory_format=torch.channels_last
+                                   ).requires_grad_(True)
+    y = m(x, r)
+    (y * gy.to("cuda:0")).sum().backward()
+
+    xr = x0.clone().requires_grad_(True)
+    rr = r0.clone().requires_grad_(True)
+    wr = m.weight.detach().cpu().clone().requires_grad_(True)
+    br = m.bias.detach().cpu().clone().requires_grad_(True)
+    yr = F.relu(F.batch_norm(xr, torch.zeros(C), torch.ones(C), wr, br,
+                             True, 0.1, 1e-5) + rr)
+    (yr * gy).sum().backward()
+
+    torch.testing.assert_close(y.cpu(), yr, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(x.grad.cpu(), xr.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(r.grad.cpu(), rr.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(m.weight.grad.cpu(), wr.grad, rtol=1e-4, atol=1e-3)
+    torch.testing.assert_close(m.bias.grad.cpu(), br.grad, rtol=1e-4, atol=1e-3)
+
+
+def test_fused_bn_eval_mode():
+    from mine_amd.ops.bn import FusedBNAct
+    import torch.nn.functional as F
+    g = torch.Generator().manual_seed(13)
+    C = 6
+    m = FusedBNAct(C, act="relu").to("cuda:0").eval()
+    with torch.no_grad():
+        m.running_mean.copy_(torch.randn(C, generator=g) * 0.3)
+        m.running_var.copy_(torch.rand(C, generator=g) + 0.5)
+        x = torch.randn(2, C, 9, 11, generator=g).to("cuda:0").contiguous(
+            memory_format=torch.channels_last)
+        y = m(x)
+        yr = F.relu(F.batch_norm(x.cpu(), m.running_mean.cpu(),
+                                 m.running_var.cpu(), m.weight.cpu(),
+                                 m.bias.cpu(), False, 0.1, 1e-5))
+    torch.testing.assert_close(y.cpu(), yr, rtol=1e-5, atol=1e-5)

@@ -131,8 +131,14 @@ class SynthesisTask:
             if bool(config.get("training.sync_batchnorm", False)) and \
                     torch.distributed.is_initialized() and \
                     torch.distributed.get_world_size() > 1:
-                self.backbone = torch.nn.SyncBatchNorm.convert_sync_batchnorm(self.backbone)
-                self.decoder = torch.nn.SyncBatchNorm.convert_sync_batchnorm(self.decoder)
+                # cross-rank BN statistics (the reference's SyncBatchNorm
+                # role, ref synthesis_task.py:106-112) via FusedBNAct's own
+                # stat all-reduce — torch's convert_sync_batchnorm would
+                # drop the fused activation epilogues.
+                from mine_amd.ops.bn import FusedBNAct
+                for m in list(self.backbone.modules()) + list(self.decoder.modules()):
+                    if isinstance(m, FusedBNAct):
+                        m.sync = True
             self.grad_engine = GradAllReduceEngine(
                 [self.backbone, self.decoder],
                 bucket_mb=float(config.get("training.grad_bucket_mb", 25)))

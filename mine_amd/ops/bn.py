@@ -28,24 +28,35 @@ _ACT = {"none": 0, "relu": 1, "lrelu": 2, "elu": 3, "add_relu": 4}
 
 class _BNActFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x_flat, res_flat, mean, invstd, gamma, beta, M, C, act):
+    def forward(ctx, x_flat, res_flat, mean, invstd, gamma, beta, M, C, act,
+                sync):
         ext = get_extension(required=True)
         y = ext.bn_act_fwd(x_flat, res_flat, mean, invstd,
                            gamma, beta, M, C, act)
         ctx.save_for_backward(x_flat, res_flat, mean, invstd, gamma, beta)
-        ctx.geom = (M, C, act)
+        ctx.geom = (M, C, act, sync)
         return y
 
     @staticmethod
     def backward(ctx, gy):
         ext = get_extension(required=True)
         x_flat, res_flat, mean, invstd, gamma, beta = ctx.saved_tensors
-        M, C, act = ctx.geom
-        dx, dres, dgamma, dbeta = ext.bn_act_bwd(
-            x_flat, res_flat, gy.contiguous(), mean, invstd, gamma, beta,
-            M, C, act)
+        M, C, act, sync = ctx.geom
+        gy = gy.contiguous()
+        red = ext.bn_act_bwd_reduce(x_flat, res_flat, gy, mean, invstd,
+                                    gamma, beta, M, C, act)
+        # SyncBN backward: dx needs the GLOBAL (dbeta, dgamma) sums and the
+        # global batch count; the returned parameter grads stay LOCAL (DDP
+        # averages them afterwards, matching torch SyncBatchNorm + DDP).
+        dbeta, dgamma = red[:C].clone(), red[C:].clone()
+        M_norm = M
+        if sync:
+            torch.distributed.all_reduce(red)
+            M_norm = M * torch.distributed.get_world_size()
+        dx, dres = ext.bn_act_bwd_dx(x_flat, res_flat, gy, mean, invstd,
+                                     gamma, beta, red, M, C, act, M_norm)
         return (dx, dres if act == _ACT["add_relu"] else None, None, None,
-                dgamma, dbeta, None, None, None)
+                dgamma, dbeta, None, None, None, None)
 
 
 def _act_eager(act: str, z: torch.Tensor) -> torch.Tensor:
@@ -69,6 +80,14 @@ class FusedBNAct(nn.BatchNorm2d):
         super().__init__(num_features, **kw)
         assert act in _ACT, act
         self.act = act
+        # Cross-rank statistics sync (the reference's SyncBatchNorm role,
+        # ref synthesis_task.py:106-112): when True and a process group is
+        # live, the per-channel (sum, sumsq) vectors are all-reduced
+        # before normalization — one 2C-float RCCL collective per BN.
+        # Enabled by SynthesisTask when training.sync_batchnorm is set;
+        # torch's convert_sync_batchnorm would silently DROP the fused
+        # activation, so it must not be used on this module.
+        self.sync = False
 
     def _fallback(self, x, res):
         xf = x.float()
@@ -100,6 +119,9 @@ class FusedBNAct(nn.BatchNorm2d):
         res_flat = res.permute(0, 2, 3, 1).reshape(-1) if res is not None \
             else torch.empty(0, device=x.device, dtype=x.dtype)
 
+        do_sync = self.training and self.sync and \
+            torch.distributed.is_initialized() and \
+            torch.distributed.get_world_size() > 1
         if self.training:
             if self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
@@ -107,18 +129,33 @@ class FusedBNAct(nn.BatchNorm2d):
                 1.0 / float(self.num_batches_tracked)
             track = self.track_running_stats and self.running_mean is not None
             empty = torch.empty(0, device=x.device, dtype=torch.float32)
-            mean, invstd = ext.bn_stats(
-                x_flat.detach(), M, C,
-                self.running_mean if track else empty,
-                self.running_var if track else empty,
-                self.eps, mom)
+            if do_sync:
+                sums = ext.bn_sums(x_flat.detach(), M, C)
+                torch.distributed.all_reduce(sums)
+                Mg = M * torch.distributed.get_world_size()
+                mean = sums[:C] / Mg
+                var = (sums[C:] / Mg - mean * mean).clamp_min_(0.0)
+                invstd = torch.rsqrt(var + self.eps)
+                if track:
+                    with torch.no_grad():
+                        self.running_mean += mom * (mean - self.running_mean)
+                        unbiased = var * Mg / max(Mg - 1, 1)
+                        self.running_var += mom * (unbiased - self.running_var)
+                mean = mean.contiguous()
+                invstd = invstd.contiguous()
+            else:
+                mean, invstd = ext.bn_stats(
+                    x_flat.detach(), M, C,
+                    self.running_mean if track else empty,
+                    self.running_var if track else empty,
+                    self.eps, mom)
         else:
             mean = self.running_mean.float()
             invstd = torch.rsqrt(self.running_var.float() + self.eps)
 
         y = _BNActFn.apply(x_flat, res_flat, mean, invstd,
                            self.weight.float(), self.bias.float(),
-                           M, C, _ACT[self.act])
+                           M, C, _ACT[self.act], do_sync)
         return y.view(B, H, W, C).permute(0, 3, 1, 2)
 
     def extra_repr(self) -> str:  # pragma: no cover

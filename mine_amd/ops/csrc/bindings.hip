@@ -203,6 +203,14 @@ at::Tensor reflect_pad_bwd(at::Tensor gout, int64_t N, int64_t H, int64_t W,
     else TORCH_CHECK(false, "bn: dtype must be f32 or bf16");     \
   } while (0)
 
+at::Tensor bn_sums(at::Tensor x, int64_t M, int64_t C) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.numel() == M * C);
+  auto sums = at::zeros({2 * C}, x.options().dtype(at::kFloat));
+  BN_DISPATCH(mine_bn_stats, x.scalar_type(), x.data_ptr(),
+              sums.data_ptr<float>(), M, (int)C, stream());
+  return sums;
+}
+
 std::vector<at::Tensor> bn_stats(at::Tensor x, int64_t M, int64_t C,
                                  at::Tensor running_mean,
                                  at::Tensor running_var, double eps,
@@ -236,31 +244,61 @@ at::Tensor bn_act_fwd(at::Tensor x, at::Tensor res, at::Tensor mean,
   return y;
 }
 
-std::vector<at::Tensor> bn_act_bwd(at::Tensor x, at::Tensor res,
-                                   at::Tensor gy, at::Tensor mean,
-                                   at::Tensor invstd, at::Tensor gamma,
-                                   at::Tensor beta, int64_t M, int64_t C,
-                                   int64_t act) {
+at::Tensor bn_act_bwd_reduce(at::Tensor x, at::Tensor res, at::Tensor gy,
+                             at::Tensor mean, at::Tensor invstd,
+                             at::Tensor gamma, at::Tensor beta, int64_t M,
+                             int64_t C, int64_t act) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && gy.is_contiguous());
-  auto f32 = x.options().dtype(at::kFloat);
-  auto red = at::zeros({2 * C}, f32);  // (dbeta, dgamma)
+  auto red = at::zeros({2 * C}, x.options().dtype(at::kFloat));
   const bool add_relu = act == 4;
   BN_DISPATCH(mine_bn_act_bwd_reduce, x.scalar_type(), x.data_ptr(),
               add_relu ? res.data_ptr() : nullptr, gy.data_ptr(),
               mean.data_ptr<float>(), invstd.data_ptr<float>(),
               gamma.data_ptr<float>(), beta.data_ptr<float>(),
               red.data_ptr<float>(), M, (int)C, (int)act, stream());
+  return red;  // (dbeta, dgamma)
+}
+
+// M iterates the LOCAL elements; M_norm is the (possibly cross-rank
+// global) batch count in the d(mean)/d(var) terms — they differ only
+// under SyncBN, where `red` has been all-reduced in between.
+std::vector<at::Tensor> bn_act_bwd_dx(at::Tensor x, at::Tensor res,
+                                      at::Tensor gy, at::Tensor mean,
+                                      at::Tensor invstd, at::Tensor gamma,
+                                      at::Tensor beta, at::Tensor red,
+                                      int64_t M, int64_t C, int64_t act,
+                                      int64_t M_norm) {
+  const bool add_relu = act == 4;
+  auto scaled = red;
+  if (M_norm != M) {
+    // the dx kernel divides by its M argument; rescale red instead of
+    // adding a second kernel parameter
+    scaled = red * ((double)M / (double)M_norm);
+  }
   auto dx = at::empty_like(x);
   auto dres = add_relu ? at::empty_like(x) : at::empty({0}, x.options());
   BN_DISPATCH(mine_bn_act_bwd_dx, x.scalar_type(), x.data_ptr(),
               add_relu ? res.data_ptr() : nullptr, gy.data_ptr(),
               mean.data_ptr<float>(), invstd.data_ptr<float>(),
               gamma.data_ptr<float>(), beta.data_ptr<float>(),
-              red.data_ptr<float>(), dx.data_ptr(),
+              scaled.contiguous().data_ptr<float>(), dx.data_ptr(),
               add_relu ? dres.data_ptr() : nullptr, M, (int)C, (int)act,
               stream());
+  return {dx, dres};
+}
+
+std::vector<at::Tensor> bn_act_bwd(at::Tensor x, at::Tensor res,
+                                   at::Tensor gy, at::Tensor mean,
+                                   at::Tensor invstd, at::Tensor gamma,
+                                   at::Tensor beta, int64_t M, int64_t C,
+                                   int64_t act) {
+  auto red = bn_act_bwd_reduce(x, res, gy, mean, invstd, gamma, beta,
+                               M, C, act);
+  auto dxr = bn_act_bwd_dx(x, res, gy, mean, invstd, gamma, beta, red,
+                           M, C, act, M);
   // dbeta = red[:C], dgamma = red[C:]
-  return {dx, dres, red.narrow(0, C, C).clone(), red.narrow(0, 0, C).clone()};
+  return {dxr[0], dxr[1], red.narrow(0, C, C).clone(),
+          red.narrow(0, 0, C).clone()};
 }
 
 // --------------------------------------------------------------------------
@@ -328,7 +366,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "gather (atomic-free) reflection pad backward");
   mod.def("bn_stats", &bn_stats,
           "per-channel mean/invstd + running-stat update");
+  mod.def("bn_sums", &bn_sums,
+          "per-channel (sum, sumsq) only — for cross-rank SyncBN");
   mod.def("bn_act_fwd", &bn_act_fwd, "fused normalize + activation");
   mod.def("bn_act_bwd", &bn_act_bwd,
           "fused BN+act backward -> dx, dres, dgamma, dbeta");
+  mod.def("bn_act_bwd_reduce", &bn_act_bwd_reduce);
+  mod.def("bn_act_bwd_dx", &bn_act_bwd_dx);
 }

@@ -207,15 +207,14 @@ class SynthesisTask:
     # ------------------------------------------------------------------
     def mpi_predictor(self, src_imgs: torch.Tensor, disparity: torch.Tensor
                       ) -> List[torch.Tensor]:
-        """Backbone + decoder -> 4 per-scale packed MPIs (B,S,H_s,W_s,4) fp32."""
+        """Backbone + decoder -> 4 per-scale packed MPIs (B,S,H_s,W_s,4) fp32.
+
+        The decoder packs via the fused head kernel (one pass; the eager
+        sigmoid/abs/cat/permute chain was a top profile entry)."""
         with self._autocast():
             feats = self.backbone(src_imgs)
-            outputs = self.decoder(feats, disparity)
-        mpis = []
-        for s in range(4):
-            mpi = outputs[("disp", s)].float()  # BxSx4xHxW
-            mpis.append(mpi.permute(0, 1, 3, 4, 2).contiguous())
-        return mpis
+            outputs = self.decoder(feats, disparity, packed=True)
+        return [outputs[("disp", s)] for s in range(4)]
 
     def network_forward(self) -> Dict[str, object]:
         """CS2: disparity sampling -> (coarse-to-fine) MPI prediction."""

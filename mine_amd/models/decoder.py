@@ -121,10 +121,14 @@ class MPIDecoder(nn.Module):
         return torch.cat((f, p), dim=1)
 
     def forward(self, input_features: List[torch.Tensor],
-                disparity: torch.Tensor) -> Dict[Tuple[str, int], torch.Tensor]:
+                disparity: torch.Tensor, packed: bool = False
+                ) -> Dict[Tuple[str, int], torch.Tensor]:
         """input_features: 5 taps BxCxHxW; disparity: BxS.
 
         Returns {("disp", s): BxSx4xH_sxW_s} with rgb=sigmoid, sigma=|x|+1e-4.
+        With ``packed=True`` each entry is instead the packed fp32
+        (B,S,H_s,W_s,4) MPI the fused renderer consumes, produced by the
+        one-pass head kernel (mine_amd/ops/head.py).
         """
         B, S = disparity.shape
         pe = self.embedder(disparity.reshape(B * S, 1))  # (B*S, E)
@@ -155,7 +159,12 @@ class MPIDecoder(nn.Module):
             if i in self.scales:
                 out = self.dispconvs[str(i)](x)
                 Hs, Ws = out.shape[-2:]
-                mpi = out.view(B, S, self.num_output_channels, Hs, Ws)
+                if packed and not (self.sigma_dropout_rate > 0.0 and self.training):
+                    from mine_amd.ops.head import mpi_head_pack
+                    outputs[("disp", i)] = mpi_head_pack(out, B, S,
+                                                         self.use_alpha)
+                    continue
+                mpi = out.float().view(B, S, self.num_output_channels, Hs, Ws)
                 rgb = torch.sigmoid(mpi[:, :, 0:3])
                 if self.use_alpha:
                     sigma = torch.sigmoid(mpi[:, :, 3:])
@@ -165,5 +174,8 @@ class MPIDecoder(nn.Module):
                     sigma = F.dropout2d(
                         sigma.view(B * S, 1, Hs, Ws), p=self.sigma_dropout_rate
                     ).view(B, S, 1, Hs, Ws)
-                outputs[("disp", i)] = torch.cat((rgb, sigma), dim=2)
+                mpi = torch.cat((rgb, sigma), dim=2)
+                if packed:
+                    mpi = mpi.permute(0, 1, 3, 4, 2).contiguous()
+                outputs[("disp", i)] = mpi
         return outputs

@@ -33,6 +33,12 @@ void mine_reflect_pad_bwd_f32(const float*, float*, int, int, int, int, int,
                               hipStream_t);
 void mine_reflect_pad_bwd_bf16(const void*, void*, int, int, int, int, int,
                                hipStream_t);
+void mine_mpi_head_fwd_f32(const void*, float*, int64_t, int, hipStream_t);
+void mine_mpi_head_fwd_bf16(const void*, float*, int64_t, int, hipStream_t);
+void mine_mpi_head_bwd_f32(const void*, const float*, void*, int64_t, int,
+                           hipStream_t);
+void mine_mpi_head_bwd_bf16(const void*, const float*, void*, int64_t, int,
+                            hipStream_t);
 void mine_bn_stats_f32(const void*, float*, int64_t, int, hipStream_t);
 void mine_bn_stats_bf16(const void*, float*, int64_t, int, hipStream_t);
 void mine_bn_finalize(const float*, float*, float*, float*, float*, int64_t,
@@ -202,6 +208,33 @@ at::Tensor reflect_pad_bwd(at::Tensor gout, int64_t N, int64_t H, int64_t W,
     else if ((T) == at::kBFloat16) FN##_bf16(__VA_ARGS__);        \
     else TORCH_CHECK(false, "bn: dtype must be f32 or bf16");     \
   } while (0)
+
+// fused MPI head over a flat (N,4) view; out is fp32 (N,4)
+at::Tensor mpi_head_fwd(at::Tensor z, int64_t N, bool alpha) {
+  TORCH_CHECK(z.is_cuda() && z.is_contiguous() && z.numel() == N * 4);
+  auto out = at::empty({N * 4}, z.options().dtype(at::kFloat));
+  if (z.scalar_type() == at::kFloat)
+    mine_mpi_head_fwd_f32(z.data_ptr(), out.data_ptr<float>(), N,
+                          alpha ? 1 : 0, stream());
+  else if (z.scalar_type() == at::kBFloat16)
+    mine_mpi_head_fwd_bf16(z.data_ptr(), out.data_ptr<float>(), N,
+                           alpha ? 1 : 0, stream());
+  else
+    TORCH_CHECK(false, "mpi_head: dtype must be f32 or bf16");
+  return out;
+}
+
+at::Tensor mpi_head_bwd(at::Tensor z, at::Tensor gout, int64_t N, bool alpha) {
+  TORCH_CHECK(gout.is_contiguous() && gout.scalar_type() == at::kFloat);
+  auto gz = at::empty_like(z);
+  if (z.scalar_type() == at::kFloat)
+    mine_mpi_head_bwd_f32(z.data_ptr(), gout.data_ptr<float>(), gz.data_ptr(),
+                          N, alpha ? 1 : 0, stream());
+  else
+    mine_mpi_head_bwd_bf16(z.data_ptr(), gout.data_ptr<float>(),
+                           gz.data_ptr(), N, alpha ? 1 : 0, stream());
+  return gz;
+}
 
 at::Tensor bn_sums(at::Tensor x, int64_t M, int64_t C) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.numel() == M * C);
@@ -373,4 +406,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused BN+act backward -> dx, dres, dgamma, dbeta");
   mod.def("bn_act_bwd_reduce", &bn_act_bwd_reduce);
   mod.def("bn_act_bwd_dx", &bn_act_bwd_dx);
+  mod.def("mpi_head_fwd", &mpi_head_fwd,
+          "dispconv out -> packed fp32 MPI (sigmoid rgb, |x|+1e-4 sigma)");
+  mod.def("mpi_head_bwd", &mpi_head_bwd);
 }

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "ssim_kernels.hip"),
         os.path.join(CSRC, "pad_kernels.hip"),
         os.path.join(CSRC, "bn_kernels.hip"),
+        os.path.join(CSRC, "head_kernels.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

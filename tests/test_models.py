@@ -58,3 +58,24 @@ def test_decoder_depends_on_disparity():
         o1 = dec([f.clone() for f in feats], torch.tensor([[0.9, 0.5]]))
         o2 = dec([f.clone() for f in feats], torch.tensor([[0.8, 0.1]]))
     assert (o1[("disp", 0)] - o2[("disp", 0)]).abs().max() > 1e-6
+
+
+def test_decoder_packed_output_matches_unpacked():
+    import torch
+    from mine_amd.models import MPIDecoder, ResNetEncoder
+
+    torch.manual_seed(0)
+    enc = ResNetEncoder(50).eval()
+    dec = MPIDecoder(num_ch_enc=enc.num_ch_enc).eval()
+    img = torch.rand(1, 3, 64, 96)
+    disp = torch.linspace(0.9, 0.1, 4).unsqueeze(0)
+    with torch.no_grad():
+        feats = enc(img)
+        out_u = dec(feats, disp)
+        out_p = dec(feats, disp, packed=True)
+    for s in range(4):
+        unpacked = out_u[("disp", s)]            # B,S,4,H,W
+        packed = out_p[("disp", s)]              # B,S,H,W,4
+        assert packed.shape == (1, 4, 64 >> s, 96 >> s, 4)
+        torch.testing.assert_close(
+            packed, unpacked.permute(0, 1, 3, 4, 2).contiguous())

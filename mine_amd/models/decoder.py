@@ -45,6 +45,58 @@ class ConvBlock(nn.Module):
         return self.bn(self.conv(self.pad(x)))
 
 
+class SplitConvBlock(nn.Module):
+    """ConvBlock over cat(x_dec, base broadcast over S, PE broadcast) —
+    WITHOUT materializing the B*S expansion.
+
+    The reference expands every skip map B -> B*S, concatenates the
+    positional encoding and convolves the (C_dec + C_enc + E)-channel
+    stack at batch B*S (ref depth_decoder.py:103-116). Because the
+    expanded channels are identical across the S planes, conv linearity
+    factors the block EXACTLY (reflection padding preserves constant
+    fields, so the PE contribution is a pure per-(b,s,k) bias):
+
+        out = conv_dec(pad(x_dec))                      # batch B*S
+            + conv_skip(pad(base))[broadcast over S]    # batch B  (S x fewer FLOPs)
+            + (pe @ sum_k3x3 W_pe^T)[:, :, None, None]  # (B*S, C_out) bias
+
+    This removes the expanded-concat materialization (GBs of copies per
+    step), runs the wide encoder channels at batch B instead of B*S, and
+    eliminates the tile-hostile 1301/661/341/2069-channel conv shapes the
+    profile showed MIOpen collapsing on (profiles/r01_progress.md). The
+    single `conv` keeps the reference's (C_out, C_dec+C_enc+E, 3, 3)
+    weight shape, so checkpoint keys and shapes are unchanged.
+    """
+
+    def __init__(self, dec_ch: int, base_ch: int, pe_ch: int, out_ch: int):
+        super().__init__()
+        from mine_amd.ops.bn import FusedBNAct
+        from mine_amd.ops.pad import ReflectionPad2d
+        self.dec_ch, self.base_ch, self.pe_ch = dec_ch, base_ch, pe_ch
+        self.pad = ReflectionPad2d(1)
+        self.conv = nn.Conv2d(dec_ch + base_ch + pe_ch, out_ch, 3)
+        self.bn = FusedBNAct(out_ch, act="elu")
+
+    def forward(self, x_dec, base, pe, B: int, S: int) -> torch.Tensor:
+        w = self.conv.weight
+        d, b = self.dec_ch, self.base_ch
+        # base part at batch B (bias lives here; it is S-invariant)
+        y_base = F.conv2d(self.pad(base), w[:, d:d + b], self.conv.bias)
+        K, Hb, Wb = y_base.shape[1:]
+        # PE part: conv of a spatially-constant field == channel bias
+        w_pe = w[:, d + b:].sum((2, 3))  # (K, E)
+        bias_pe = torch.matmul(pe.to(w_pe.dtype), w_pe.t())  # (B*S, K)
+        # broadcast-add in NHWC so the (B*S,K,H,W) result is channels_last
+        # without a transpose (conv outputs are NHWC on GPU already)
+        yb = y_base.permute(0, 2, 3, 1).unsqueeze(1) \
+            + bias_pe.view(B, S, 1, 1, K).to(y_base.dtype)
+        yb = yb.view(B * S, Hb, Wb, K)
+        if d:
+            y_dec = F.conv2d(self.pad(x_dec), w[:, :d], None)
+            yb = yb + y_dec.permute(0, 2, 3, 1)
+        return self.bn(yb.permute(0, 3, 1, 2))
+
+
 def _neck_conv(in_ch: int, out_ch: int, k: int) -> nn.Sequential:
     from mine_amd.ops.bn import FusedBNAct
     return nn.Sequential(
@@ -94,27 +146,44 @@ class MPIDecoder(nn.Module):
         self.conv_up1 = _neck_conv(256, 256, 3)
         self.conv_up2 = _neck_conv(256, c_last, 1)
 
-        # channels after PE concat
-        ch_enc = [c + E for c in num_ch_enc]
+        # channels after PE concat (weight shapes keep the reference's
+        # concat layout even though the forward factors the concat away)
         ch_dec = [16, 32, 64, 128, 256]
+        self.num_ch_enc = list(num_ch_enc)
 
         self.upconvs0 = nn.ModuleList()
         self.upconvs1 = nn.ModuleList()
         for i in range(4, -1, -1):
-            c_in = ch_enc[-1] if i == 4 else ch_dec[i + 1]
-            self.upconvs0.append(ConvBlock(c_in, ch_dec[i]))
-            c_in = ch_dec[i]
+            if i == 4:
+                self.upconvs0.append(
+                    SplitConvBlock(0, num_ch_enc[-1], E, ch_dec[i]))
+            else:
+                self.upconvs0.append(ConvBlock(ch_dec[i + 1], ch_dec[i]))
             if use_skips and i > 0:
-                c_in += ch_enc[i - 1]
-            self.upconvs1.append(ConvBlock(c_in, ch_dec[i]))
+                self.upconvs1.append(
+                    SplitConvBlock(ch_dec[i], num_ch_enc[i - 1], E, ch_dec[i]))
+            else:
+                self.upconvs1.append(ConvBlock(ch_dec[i], ch_dec[i]))
         self.dispconvs = nn.ModuleDict({
             str(s): Conv3x3Refl(ch_dec[s], num_output_channels) for s in self.scales
         })
 
+    def _up(self, x: torch.Tensor) -> torch.Tensor:
+        """Nearest x2 upsample, autocast-shielded: CUDA autocast promotes
+        upsample_nearest to fp32, which poisoned the downstream convs/pads
+        into fp32 (observed in profiles); outside autocast the op keeps
+        the tensor's own dtype."""
+        if x.is_cuda and torch.is_autocast_enabled():
+            with torch.autocast("cuda", enabled=False):
+                return self.upsample(x)
+        return self.upsample(x)
+
     def _expand_with_pe(self, feat: torch.Tensor, pe: torch.Tensor,
                         B: int, S: int) -> torch.Tensor:
         """feat: BxCxHxW -> (B*S)x(C+E)xHxW with the plane's PE broadcast
-        over HxW (ref depth_decoder.py:103-116)."""
+        over HxW — the reference's materialized expansion (ref
+        depth_decoder.py:103-116). Kept as the oracle for
+        SplitConvBlock's factored equivalent; not used in forward."""
         _, C, H, W = feat.shape
         f = feat.unsqueeze(1).expand(B, S, C, H, W).reshape(B * S, C, H, W)
         p = pe.to(feat.dtype).unsqueeze(-1).unsqueeze(-1).expand(B * S, self.E, H, W)
@@ -137,25 +206,27 @@ class MPIDecoder(nn.Module):
         enc_out = input_features[-1]
         x = self.conv_down1(self.downsample(enc_out))
         x = self.conv_down2(self.downsample(x))
-        x = self.conv_up1(self.upsample(x))
-        neck = self.conv_up2(self.upsample(x))  # BxC_lastxH/32xW/32
-
-        skips = [self._expand_with_pe(f, pe, B, S) for f in input_features]
-        x = self._expand_with_pe(neck, pe, B, S)
+        x = self.conv_up1(self._up(x))
+        neck = self.conv_up2(self._up(x))  # BxC_lastxH/32xW/32
 
         outputs: Dict[Tuple[str, int], torch.Tensor] = {}
         for idx, i in enumerate(range(4, -1, -1)):
-            x = self.upconvs0[idx](x)
-            x = self.upsample(x)
+            if i == 4:
+                # B -> B*S happens HERE, factored: neck at batch B + PE bias
+                x = self.upconvs0[idx](None, neck, pe, B, S)
+            else:
+                x = self.upconvs0[idx](x)
+            x = self._up(x)
             if self.use_skips and i > 0:
-                skip = skips[i - 1]
+                skip = input_features[i - 1]  # RAW batch-B tap
                 if x.shape[-2:] != skip.shape[-2:]:
                     # odd intermediate sizes (e.g. H/32 == 3): align to the
                     # skip tap (the reference decoder requires power-of-two
                     # -divisible sizes and would crash here)
                     x = F.interpolate(x, size=skip.shape[-2:], mode="nearest")
-                x = torch.cat((x, skip), dim=1)
-            x = self.upconvs1[idx](x)
+                x = self.upconvs1[idx](x, skip, pe, B, S)
+            else:
+                x = self.upconvs1[idx](x)
             if i in self.scales:
                 out = self.dispconvs[str(i)](x)
                 Hs, Ws = out.shape[-2:]

@@ -79,3 +79,45 @@ def test_decoder_packed_output_matches_unpacked():
         assert packed.shape == (1, 4, 64 >> s, 96 >> s, 4)
         torch.testing.assert_close(
             packed, unpacked.permute(0, 1, 3, 4, 2).contiguous())
+
+
+def test_split_conv_block_matches_materialized_concat():
+    """SplitConvBlock's factored conv == conv over the expanded concat
+    (exactness of the B->B*S factorization; reflection pad preserves
+    constant PE fields)."""
+    import torch
+    import torch.nn.functional as F
+    from mine_amd.models.decoder import SplitConvBlock
+
+    torch.manual_seed(5)
+    B, S, E = 2, 3, 21
+    dec_ch, base_ch, out_ch, H, W = 8, 12, 6, 10, 14
+    blk = SplitConvBlock(dec_ch, base_ch, E, out_ch).eval()
+
+    x_dec = torch.randn(B * S, dec_ch, H, W)
+    base = torch.randn(B, base_ch, H, W)
+    pe = torch.randn(B * S, E)
+
+    with torch.no_grad():
+        y = blk(x_dec, base, pe, B, S)
+
+        # oracle: materialize expand + concat, one conv, same bn
+        base_x = base.unsqueeze(1).expand(B, S, base_ch, H, W
+                                          ).reshape(B * S, base_ch, H, W)
+        pe_x = pe[:, :, None, None].expand(B * S, E, H, W)
+        cat = torch.cat((x_dec, base_x, pe_x), dim=1)
+        z = F.conv2d(F.pad(cat, (1, 1, 1, 1), mode="reflect"),
+                     blk.conv.weight, blk.conv.bias)
+        ref = blk.bn(z)
+
+    torch.testing.assert_close(y, ref, rtol=1e-4, atol=1e-5)
+
+    # no-dec variant (the neck stage)
+    blk0 = SplitConvBlock(0, base_ch, E, out_ch).eval()
+    with torch.no_grad():
+        y0 = blk0(None, base, pe, B, S)
+        cat0 = torch.cat((base_x, pe_x), dim=1)
+        z0 = F.conv2d(F.pad(cat0, (1, 1, 1, 1), mode="reflect"),
+                      blk0.conv.weight, blk0.conv.bias)
+        ref0 = blk0.bn(z0)
+    torch.testing.assert_close(y0, ref0, rtol=1e-4, atol=1e-5)

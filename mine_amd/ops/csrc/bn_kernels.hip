@@ -55,25 +55,33 @@ __device__ __forceinline__ float act_grad(int act, float z) {
 // stats: per-channel sum and sum-of-squares over the N*H*W axis
 // ---------------------------------------------------------------------------
 
+// channel-group width: the smallest power of two >= min(C, 64), so
+// small-C layers (the decoder's C=16 blocks) keep every lane busy
+// instead of idling 3/4 of the block.
+inline __host__ __device__ int chan_group(int C) {
+  int g = 1;
+  while (g < C && g < 64) g <<= 1;
+  return g;
+}
+
 template <typename T>
 __global__ void __launch_bounds__(kBlock)
 bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sums,  // (2,C)
-                int64_t M, int C) {
-  // grid.x: slabs of the M axis; grid.y: channel chunks of 64
-  const int c0 = blockIdx.y * 64;
-  const int nc = min(64, C - c0);
+                int64_t M, int C, int cg) {
+  // grid.x: slabs of the M axis; grid.y: channel chunks of cg
+  const int c0 = blockIdx.y * cg;
+  const int nc = min(cg, C - c0);
   __shared__ float s_sum[64], s_sq[64];
-  for (int i = threadIdx.x; i < 64; i += kBlock) {
+  for (int i = threadIdx.x; i < cg; i += kBlock) {
     s_sum[i] = 0.0f;
     s_sq[i] = 0.0f;
   }
   __syncthreads();
 
-  // each thread owns channel (c0 + tid%64) and strides rows by
-  // (kBlock/64 * gridDim.x)
-  const int lane_c = threadIdx.x & 63;
-  const int row0 = blockIdx.x * (kBlock / 64) + (threadIdx.x >> 6);
-  const int rstride = gridDim.x * (kBlock / 64);
+  const int lane_c = threadIdx.x & (cg - 1);
+  const int rows_per_blk = kBlock / cg;
+  const int row0 = blockIdx.x * rows_per_blk + threadIdx.x / cg;
+  const int rstride = gridDim.x * rows_per_blk;
   float lsum = 0.0f, lsq = 0.0f;
   if (lane_c < nc) {
     const int c = c0 + lane_c;
@@ -154,18 +162,19 @@ bn_act_bwd_reduce_kernel(const T* __restrict__ x, const T* __restrict__ res,
                          const float* __restrict__ gamma,
                          const float* __restrict__ beta,
                          float* __restrict__ out,  // (2,C): dbeta, dgamma
-                         int64_t M, int C, int act) {
-  const int c0 = blockIdx.y * 64;
-  const int nc = min(64, C - c0);
+                         int64_t M, int C, int act, int cg) {
+  const int c0 = blockIdx.y * cg;
+  const int nc = min(cg, C - c0);
   __shared__ float s_db[64], s_dg[64];
-  for (int i = threadIdx.x; i < 64; i += kBlock) {
+  for (int i = threadIdx.x; i < cg; i += kBlock) {
     s_db[i] = 0.0f;
     s_dg[i] = 0.0f;
   }
   __syncthreads();
-  const int lane_c = threadIdx.x & 63;
-  const int row0 = blockIdx.x * (kBlock / 64) + (threadIdx.x >> 6);
-  const int rstride = gridDim.x * (kBlock / 64);
+  const int lane_c = threadIdx.x & (cg - 1);
+  const int rows_per_blk = kBlock / cg;
+  const int row0 = blockIdx.x * rows_per_blk + threadIdx.x / cg;
+  const int rstride = gridDim.x * rows_per_blk;
   if (lane_c < nc) {
     const int c = c0 + lane_c;
     const float mu = mean[c], is = invstd[c], ga = gamma[c], be = beta[c];
@@ -223,12 +232,13 @@ inline int grid_elems(int64_t total) {
   return (int)(g < 65535 ? g : 65535);
 }
 
-inline dim3 grid_reduce(int64_t M, int C) {
+inline dim3 grid_reduce(int64_t M, int C, int cg) {
   // enough slabs to fill 256 CUs x a few blocks, bounded
-  int64_t rows = (M + (kBlock / 64) - 1) / (kBlock / 64);
+  const int rows_per_blk = kBlock / cg;
+  int64_t rows = (M + rows_per_blk - 1) / rows_per_blk;
   int gx = (int)(rows < 2048 ? rows : 2048);
   if (gx < 1) gx = 1;
-  return dim3(gx, (C + 63) / 64);
+  return dim3(gx, (C + cg - 1) / cg);
 }
 
 }  // namespace
@@ -236,8 +246,10 @@ inline dim3 grid_reduce(int64_t M, int C) {
 #define EXPORT_BN(SUF, T)                                                      \
   extern "C" void mine_bn_stats_##SUF(const void* x, float* sums, int64_t M,   \
                                       int C, hipStream_t s) {                  \
-    hipLaunchKernelGGL(bn_stats_kernel<T>, grid_reduce(M, C), dim3(kBlock), 0, \
-                       s, reinterpret_cast<const T*>(x), sums, M, C);          \
+    const int cg = chan_group(C);                                              \
+    hipLaunchKernelGGL(bn_stats_kernel<T>, grid_reduce(M, C, cg),              \
+                       dim3(kBlock), 0, s, reinterpret_cast<const T*>(x),      \
+                       sums, M, C, cg);                                        \
   }                                                                            \
   extern "C" void mine_bn_act_fwd_##SUF(                                       \
       const void* x, const void* res, const float* mean, const float* invstd,  \
@@ -252,11 +264,12 @@ inline dim3 grid_reduce(int64_t M, int C) {
       const void* x, const void* res, const void* gy, const float* mean,       \
       const float* invstd, const float* gamma, const float* beta, float* out,  \
       int64_t M, int C, int act, hipStream_t s) {                              \
-    hipLaunchKernelGGL(bn_act_bwd_reduce_kernel<T>, grid_reduce(M, C),         \
+    const int cg = chan_group(C);                                              \
+    hipLaunchKernelGGL(bn_act_bwd_reduce_kernel<T>, grid_reduce(M, C, cg),     \
                        dim3(kBlock), 0, s, reinterpret_cast<const T*>(x),      \
                        reinterpret_cast<const T*>(res),                        \
                        reinterpret_cast<const T*>(gy), mean, invstd, gamma,    \
-                       beta, out, M, C, act);                                  \
+                       beta, out, M, C, act, cg);                              \
   }                                                                            \
   extern "C" void mine_bn_act_bwd_dx_##SUF(                                    \
       const void* x, const void* res, const void* gy, const float* mean,       \

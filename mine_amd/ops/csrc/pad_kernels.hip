@@ -23,16 +23,25 @@ namespace {
 
 constexpr int kBlock = 256;
 
-template <typename T>
+// V consecutive channels per thread (one 16-byte load/store when
+// V*sizeof(T) == 16 and C % V == 0)
+template <typename T, int V>
+struct alignas(sizeof(T) * V) VecT {
+  T v[V];
+};
+
+template <typename T, int V>
 __global__ void __launch_bounds__(kBlock)
 reflect_pad_fwd_kernel(const T* __restrict__ in, T* __restrict__ out,
                        int N, int H, int W, int C, int pad) {
+  using Vec = VecT<T, V>;
+  const int Cv = C / V;
   const int Ho = H + 2 * pad, Wo = W + 2 * pad;
-  const int64_t total = (int64_t)N * Ho * Wo * C;
+  const int64_t total = (int64_t)N * Ho * Wo * Cv;
   for (int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * kBlock) {
-    const int c = (int)(i % C);
-    int64_t r = i / C;
+    const int c = (int)(i % Cv);
+    int64_t r = i / Cv;
     const int xo = (int)(r % Wo);
     r /= Wo;
     const int yo = (int)(r % Ho);
@@ -43,20 +52,23 @@ reflect_pad_fwd_kernel(const T* __restrict__ in, T* __restrict__ out,
     int xi = xo - pad;
     if (xi < 0) xi = -xi;
     if (xi >= W) xi = 2 * (W - 1) - xi;
-    out[i] = in[(((int64_t)n * H + yi) * W + xi) * C + c];
+    reinterpret_cast<Vec*>(out)[i] = reinterpret_cast<const Vec*>(
+        in)[(((int64_t)n * H + yi) * W + xi) * Cv + c];
   }
 }
 
-template <typename T>
+template <typename T, int V>
 __global__ void __launch_bounds__(kBlock)
 reflect_pad_bwd_kernel(const T* __restrict__ gout, T* __restrict__ gin,
                        int N, int H, int W, int C, int pad) {
+  using Vec = VecT<T, V>;
+  const int Cv = C / V;
   const int Ho = H + 2 * pad, Wo = W + 2 * pad;
-  const int64_t total = (int64_t)N * H * W * C;
+  const int64_t total = (int64_t)N * H * W * Cv;
   for (int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * kBlock) {
-    const int c = (int)(i % C);
-    int64_t r = i / C;
+    const int c = (int)(i % Cv);
+    int64_t r = i / Cv;
     const int xi = (int)(r % W);
     r /= W;
     const int yi = (int)(r % H);
@@ -71,12 +83,21 @@ reflect_pad_bwd_kernel(const T* __restrict__ gout, T* __restrict__ gin,
     if (xi >= 1 && xi <= pad) xs[nx++] = pad - xi;
     if (xi >= W - 1 - pad && xi <= W - 2) xs[nx++] = 2 * (W - 1) - xi + pad;
 
-    float acc = 0.0f;
-    const T* gb = gout + (int64_t)n * Ho * Wo * C + c;
+    float acc[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) acc[j] = 0.0f;
+    const Vec* gb = reinterpret_cast<const Vec*>(gout) +
+                    (int64_t)n * Ho * Wo * Cv + c;
     for (int a = 0; a < ny; ++a)
-      for (int b = 0; b < nx; ++b)
-        acc += (float)gb[((int64_t)ys[a] * Wo + xs[b]) * C];
-    gin[i] = (T)acc;
+      for (int b = 0; b < nx; ++b) {
+        const Vec g = gb[((int64_t)ys[a] * Wo + xs[b]) * Cv];
+#pragma unroll
+        for (int j = 0; j < V; ++j) acc[j] += (float)g.v[j];
+      }
+    Vec o;
+#pragma unroll
+    for (int j = 0; j < V; ++j) o.v[j] = (T)acc[j];
+    reinterpret_cast<Vec*>(gin)[i] = o;
   }
 }
 
@@ -85,40 +106,62 @@ inline int grid_for(int64_t total) {
   return (int)(g < 65535 ? g : 65535);
 }
 
+// widest V with V*sizeof(T)<=16 dividing C
+template <typename T>
+inline int pick_vec(int C) {
+  const int vmax = 16 / (int)sizeof(T);
+  for (int v = vmax; v > 1; v >>= 1)
+    if (C % v == 0) return v;
+  return 1;
+}
+
 }  // namespace
+
+#define PAD_LAUNCH(KERN, T, TOTAL, ...)                                       \
+  do {                                                                        \
+    const int v = pick_vec<T>(C);                                             \
+    const int64_t tv = (TOTAL) / v;                                           \
+    if (v == 8)                                                               \
+      hipLaunchKernelGGL((KERN<T, 8>), dim3(grid_for(tv)), dim3(kBlock), 0,   \
+                         stream, __VA_ARGS__);                                \
+    else if (v == 4)                                                          \
+      hipLaunchKernelGGL((KERN<T, 4>), dim3(grid_for(tv)), dim3(kBlock), 0,   \
+                         stream, __VA_ARGS__);                                \
+    else if (v == 2)                                                          \
+      hipLaunchKernelGGL((KERN<T, 2>), dim3(grid_for(tv)), dim3(kBlock), 0,   \
+                         stream, __VA_ARGS__);                                \
+    else                                                                      \
+      hipLaunchKernelGGL((KERN<T, 1>), dim3(grid_for(tv)), dim3(kBlock), 0,   \
+                         stream, __VA_ARGS__);                                \
+  } while (0)
 
 extern "C" {
 
 void mine_reflect_pad_fwd_f32(const float* in, float* out, int N, int H,
                               int W, int C, int pad, hipStream_t stream) {
-  hipLaunchKernelGGL(reflect_pad_fwd_kernel<float>,
-                     dim3(grid_for((int64_t)N * (H + 2 * pad) * (W + 2 * pad) * C)),
-                     dim3(kBlock), 0, stream, in, out, N, H, W, C, pad);
+  PAD_LAUNCH(reflect_pad_fwd_kernel, float,
+             (int64_t)N * (H + 2 * pad) * (W + 2 * pad) * C,
+             in, out, N, H, W, C, pad);
 }
 
 void mine_reflect_pad_fwd_bf16(const void* in, void* out, int N, int H,
                                int W, int C, int pad, hipStream_t stream) {
-  hipLaunchKernelGGL(reflect_pad_fwd_kernel<__hip_bfloat16>,
-                     dim3(grid_for((int64_t)N * (H + 2 * pad) * (W + 2 * pad) * C)),
-                     dim3(kBlock), 0, stream,
-                     reinterpret_cast<const __hip_bfloat16*>(in),
-                     reinterpret_cast<__hip_bfloat16*>(out), N, H, W, C, pad);
+  PAD_LAUNCH(reflect_pad_fwd_kernel, __hip_bfloat16,
+             (int64_t)N * (H + 2 * pad) * (W + 2 * pad) * C,
+             reinterpret_cast<const __hip_bfloat16*>(in),
+             reinterpret_cast<__hip_bfloat16*>(out), N, H, W, C, pad);
 }
 
 void mine_reflect_pad_bwd_f32(const float* gout, float* gin, int N, int H,
                               int W, int C, int pad, hipStream_t stream) {
-  hipLaunchKernelGGL(reflect_pad_bwd_kernel<float>,
-                     dim3(grid_for((int64_t)N * H * W * C)),
-                     dim3(kBlock), 0, stream, gout, gin, N, H, W, C, pad);
+  PAD_LAUNCH(reflect_pad_bwd_kernel, float, (int64_t)N * H * W * C,
+             gout, gin, N, H, W, C, pad);
 }
 
 void mine_reflect_pad_bwd_bf16(const void* gout, void* gin, int N, int H,
                                int W, int C, int pad, hipStream_t stream) {
-  hipLaunchKernelGGL(reflect_pad_bwd_kernel<__hip_bfloat16>,
-                     dim3(grid_for((int64_t)N * H * W * C)),
-                     dim3(kBlock), 0, stream,
-                     reinterpret_cast<const __hip_bfloat16*>(gout),
-                     reinterpret_cast<__hip_bfloat16*>(gin), N, H, W, C, pad);
+  PAD_LAUNCH(reflect_pad_bwd_kernel, __hip_bfloat16, (int64_t)N * H * W * C,
+             gout, gin, N, H, W, C, pad);
 }
 
 }  // extern "C"

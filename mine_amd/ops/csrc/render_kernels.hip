@@ -397,7 +397,8 @@ tgt_composite_fwd_kernel(const float* __restrict__ mpi,
     PlaneSample cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv,
                                    x, y, W, H, nullptr);
     mask += cur.inb;
-    for (int s = 0; s < S; ++s) {
+    int s = 0;
+    for (; s < S; ++s) {
       float delta;
       PlaneSample nxt;
       if (s + 1 < S) {
@@ -417,6 +418,22 @@ tgt_composite_fwd_kernel(const float* __restrict__ mpi,
       Nsum += w * cur.v.z;
       A *= (t + 1e-6f);
       cur = nxt;
+      if (A < 1e-14f) {  // transmittance dead: remaining planes add ~0
+        ++s;
+        break;
+      }
+    }
+    // mask still counts EVERY plane's validity (ref mpi_rendering.py:239):
+    // projection-only tail, no MPI gathers
+    for (int s2 = s + 1; s2 < S; ++s2) {
+      const float* Hrow = s_geom + s2 * 9;
+      const float hx = Hrow[0] * x + Hrow[1] * y + Hrow[2];
+      const float hy = Hrow[3] * x + Hrow[4] * y + Hrow[5];
+      const float hz = Hrow[6] * x + Hrow[7] * y + Hrow[8];
+      const float iz = 1.0f / hz;
+      const float u = hx * iz, v = hy * iz;
+      mask += (u > -1.0f && u < (float)W && v > -1.0f && v < (float)H)
+                  ? 1.0f : 0.0f;
     }
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
                            : (Nsum / (Wsum + 1e-5f));
@@ -519,6 +536,7 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       Nsum += w * cur.v.z;
       A *= (t + 1e-6f);
       cur = nxt;
+      if (A < 1e-14f) break;  // dead transmittance: tail adds ~0
     }
     const float Wp = Wsum + 1e-5f;
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
@@ -546,6 +564,7 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       TotalP += (double)((1.0f - te.x) * te.y) * Ad;
       Ad *= (double)(te.x + 1e-6f);
       cur = nxt;
+      if (Ad < 1e-14) break;
     }
 
     // ---- pass 3: emit gradients, bilinear scatter ----
@@ -582,6 +601,8 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       Ad *= (double)u;
       cur = nxt;
       tap = ntap;
+      // grad_mpi is pre-zeroed; dead-transmittance tail scatters ~0
+      if (Ad < 1e-14) break;
     }
   }
 }

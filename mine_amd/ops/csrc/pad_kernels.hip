@@ -161,7 +161,8 @@ void mine_reflect_pad_bwd_f32(const float* gout, float* gin, int N, int H,
 void mine_reflect_pad_bwd_bf16(const void* gout, void* gin, int N, int H,
                                int W, int C, int pad, hipStream_t stream) {
   PAD_LAUNCH(reflect_pad_bwd_kernel, __hip_bfloat16, (int64_t)N * H * W * C,
-             gout, gin, N, H, W, C, pad);
+             reinterpret_cast<const __hip_bfloat16*>(gout),
+             reinterpret_cast<__hip_bfloat16*>(gin), N, H, W, C, pad);
 }
 
 }  // extern "C"

@@ -98,6 +98,10 @@ class SynthesisTask:
         amp = str(config.get("training.amp_dtype", "bf16")).lower()
         self.amp_dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
                           "fp32": None, "float32": None}[amp]
+        # fp16 needs loss scaling (bf16/fp32 do not; driver config 5 runs
+        # Flowers in fp16)
+        self.grad_scaler = torch.amp.GradScaler("cuda") \
+            if (self.is_gpu and self.amp_dtype == torch.float16) else None
         self.channels_last = bool(config.get("training.channels_last", True)) and self.is_gpu
 
         # ---- models -----------------------------------------------------
@@ -430,11 +434,19 @@ class SynthesisTask:
             self.grad_engine.zero_grad()
         else:
             self.optimizer.zero_grad(set_to_none=False)
-        loss_dict["loss"].backward()
-        if self.grad_engine is not None:
-            self.grad_engine.finish_step()
-        mark("backward")
-        self.optimizer.step()
+        if self.grad_scaler is not None:
+            self.grad_scaler.scale(loss_dict["loss"]).backward()
+            if self.grad_engine is not None:
+                self.grad_engine.finish_step()
+            mark("backward")
+            self.grad_scaler.step(self.optimizer)
+            self.grad_scaler.update()
+        else:
+            loss_dict["loss"].backward()
+            if self.grad_engine is not None:
+                self.grad_engine.finish_step()
+            mark("backward")
+            self.optimizer.step()
         mark("optimizer")
         return loss_dict
 

@@ -60,3 +60,13 @@ def setup_logger(name: str = "mine_amd", log_file: Optional[str] = None,
     logger.setLevel(level)
     logger.propagate = False
     return logger
+
+
+def linspace_batch(start: torch.Tensor, end: torch.Tensor,
+                   steps: int) -> torch.Tensor:
+    """Batched linspace: start/end (B,) -> (B, steps)
+    (ref utils.py:70-93, which looped; this is one broadcast)."""
+    assert start.shape == end.shape and start.dim() == 1
+    t = torch.linspace(0.0, 1.0, steps, dtype=start.dtype,
+                       device=start.device)
+    return start.unsqueeze(1) + (end - start).unsqueeze(1) * t.unsqueeze(0)

@@ -65,8 +65,8 @@ def transformation_from_parameters(axisangle: torch.Tensor,
         t = t * -1
     T = torch.zeros_like(R)
     T[:, 0, 0] = T[:, 1, 1] = T[:, 2, 2] = T[:, 3, 3] = 1.0
-    T[:, :3, 3] = t.squeeze(1) if not invert else \
-        torch.matmul(R[:, :3, :3], t.squeeze(1).unsqueeze(-1)).squeeze(-1)
+    T[:, :3, 3] = t.squeeze(1)
+    # invert: R^T @ T(-t) == (T(t) @ R)^-1
     return torch.matmul(R, T) if invert else torch.matmul(T, R)
 
 

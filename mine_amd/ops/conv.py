@@ -1,0 +1,114 @@
+"""Fused reflection-pad + 3x3 conv on the MFMA kernel.
+
+Forward runs the hand-written v_mfma_f32_16x16x32_bf16 kernel
+(ops/csrc/conv_kernels.hip) with the pad folded into the LDS stage — no
+padded tensor is ever materialized. Backward materializes the pad once
+(HIP pad kernel), reuses MIOpen's tuned convolution_backward on it, and
+folds the pad gradient back with the atomic-free gather.
+
+Weights are re-packed to the exact MFMA fragment order each call (a
+cached index gather over the ~KxCx9 elements — microseconds); the
+fragment maps were measured on gfx950 with tools/mfma_probe.hip.
+"""
+from __future__ import annotations
+
+from typing import Dict, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from mine_amd.ops.backend import get_extension
+
+_LUT_CACHE: Dict[Tuple[int, int, int, torch.device], torch.Tensor] = {}
+
+
+def _pack_lut(K: int, C: int, flip: bool, device) -> torch.Tensor:
+    """Index LUT: fragment-ordered gather over a flattened (K,C,3,3)
+    weight (+one trailing zero slot for padding)."""
+    key = (K, C, flip, device)
+    lut = _LUT_CACHE.get(key)
+    if lut is not None:
+        return lut
+    Cv = C // 8
+    nseg = 9 * Cv
+    nchunks = (nseg + 3) // 4
+    nK = (K + 15) // 16
+    zero_slot = K * C * 9  # one-past-the-end: zero pad
+    idx = torch.full((nK, nchunks, 64, 8), zero_slot, dtype=torch.long)
+    for nc in range(nK):
+        for kc in range(nchunks):
+            for lane in range(64):
+                j = lane & 15
+                seg = kc * 4 + (lane >> 4)
+                kout = nc * 16 + j
+                if seg >= nseg or kout >= K:
+                    continue
+                cb, tap = divmod(seg, 9)
+                dy, dx = divmod(tap, 3)
+                if flip:
+                    dy, dx = 2 - dy, 2 - dx
+                for e in range(8):
+                    c = cb * 8 + e
+                    idx[nc, kc, lane, e] = ((kout * C + c) * 3 + dy) * 3 + dx
+    lut = idx.reshape(-1).to(device)
+    _LUT_CACHE[key] = lut
+    return lut
+
+
+def pack_weights(w: torch.Tensor, flip: bool = False) -> torch.Tensor:
+    """(K, C, 3, 3) -> fragment-ordered bf16 buffer."""
+    K, C = w.shape[0], w.shape[1]
+    lut = _pack_lut(K, C, flip, w.device)
+    flat = torch.cat((w.reshape(-1), w.new_zeros(1))).to(torch.bfloat16)
+    return flat[lut].contiguous()
+
+
+class _Conv3x3ReflFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        ext = get_extension(required=True)
+        B, C, H, W = x.shape
+        K = w.shape[0]
+        wb = w.to(torch.bfloat16)
+        wp = pack_weights(wb)
+        x_flat = x.permute(0, 2, 3, 1).reshape(-1)  # NHWC view
+        out = torch.empty(B * H * W * K, device=x.device, dtype=x.dtype)
+        ext.conv3x3_fwd(x_flat, wp,
+                        bias.float() if bias is not None else
+                        torch.empty(0, device=x.device, dtype=torch.float32),
+                        out, B, H, W, C, K, 0)
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        return out.view(B, H, W, K).permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, gy):
+        from mine_amd.ops.pad import _ReflectPadFn, reflection_pad2d
+        x, w = ctx.saved_tensors
+        with torch.no_grad():
+            xp = reflection_pad2d(x, 1)
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        gx_pad, gw, gb = torch.ops.aten.convolution_backward(
+            gy, xp, w.to(xp.dtype), [w.shape[0]] if ctx.has_bias else None,
+            [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+            [True, True, ctx.has_bias])
+        # fold the pad gradient back (atomic-free gather kernel)
+        ext = get_extension(required=True)
+        B, C, H, W = x.shape
+        flat = gx_pad.permute(0, 2, 3, 1).contiguous().reshape(-1)
+        gx = ext.reflect_pad_bwd(flat, B, H, W, C, 1)
+        gx = gx.view(B, H, W, C).permute(0, 3, 1, 2)
+        return gx, gw.to(w.dtype), (gb.to(w.dtype) if ctx.has_bias else None)
+
+
+def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
+                    bias: torch.Tensor = None) -> torch.Tensor:
+    """Reflection-pad(1) + 3x3 stride-1 conv. MFMA fast path on GPU
+    bf16 channels_last with C % 8 == 0; eager fallback otherwise."""
+    usable = (x.is_cuda and x.dtype == torch.bfloat16
+              and x.shape[1] % 8 == 0
+              and x.is_contiguous(memory_format=torch.channels_last))
+    if usable:
+        return _Conv3x3ReflFn.apply(x, w, bias)
+    return F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"),
+                    w.to(x.dtype), bias.to(x.dtype) if bias is not None else None)

@@ -33,6 +33,8 @@ void mine_reflect_pad_bwd_f32(const float*, float*, int, int, int, int, int,
                               hipStream_t);
 void mine_reflect_pad_bwd_bf16(const void*, void*, int, int, int, int, int,
                                hipStream_t);
+void mine_conv3x3_fwd(const void*, const void*, const float*, void*, int,
+                      int, int, int, int, int, hipStream_t);
 void mine_mpi_head_fwd_f32(const void*, float*, int64_t, int, hipStream_t);
 void mine_mpi_head_fwd_bf16(const void*, float*, int64_t, int, hipStream_t);
 void mine_mpi_head_bwd_f32(const void*, const float*, void*, int64_t, int,
@@ -208,6 +210,22 @@ at::Tensor reflect_pad_bwd(at::Tensor gout, int64_t N, int64_t H, int64_t W,
     else if ((T) == at::kBFloat16) FN##_bf16(__VA_ARGS__);        \
     else TORCH_CHECK(false, "bn: dtype must be f32 or bf16");     \
   } while (0)
+
+// fused reflect-pad + 3x3 conv (MFMA); flat NHWC views, bf16
+void conv3x3_fwd(at::Tensor x_flat, at::Tensor wp, at::Tensor bias,
+                 at::Tensor out, int64_t N, int64_t H, int64_t W, int64_t C,
+                 int64_t K, int64_t pad_mode) {
+  TORCH_CHECK(x_flat.is_cuda() && x_flat.is_contiguous());
+  TORCH_CHECK(x_flat.scalar_type() == at::kBFloat16 &&
+              wp.scalar_type() == at::kBFloat16 &&
+              out.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(C % 8 == 0 && x_flat.numel() == N * H * W * C);
+  TORCH_CHECK(out.numel() == N * H * W * K);
+  mine_conv3x3_fwd(x_flat.data_ptr(), wp.data_ptr(),
+                   bias.numel() ? bias.data_ptr<float>() : nullptr,
+                   out.data_ptr(), (int)N, (int)H, (int)W, (int)C, (int)K,
+                   (int)pad_mode, stream());
+}
 
 // fused MPI head over a flat (N,4) view; out is fp32 (N,4)
 at::Tensor mpi_head_fwd(at::Tensor z, int64_t N, bool alpha) {
@@ -409,4 +427,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("mpi_head_fwd", &mpi_head_fwd,
           "dispconv out -> packed fp32 MPI (sigmoid rgb, |x|+1e-4 sigma)");
   mod.def("mpi_head_bwd", &mpi_head_bwd);
+  mod.def("conv3x3_fwd", &conv3x3_fwd,
+          "fused reflect/zero-pad + 3x3 conv on MFMA 16x16x32 tiles");
 }

@@ -1,0 +1,173 @@
+// Fused reflection-pad + 3x3 stride-1 conv, NHWC bf16, MFMA 16x16x32.
+//
+// The decoder's hot remaining convs are small-channel 3x3 blocks at
+// full resolution (16->16, 16->4, 32->16... at (B*S)=256 batch); MIOpen
+// reaches ~86 TF/s on them and needs the pad materialized. This kernel
+// folds the reflection pad into the LDS stage (no padded tensor ever
+// exists) and runs the GEMM on v_mfma_f32_16x16x32_bf16 tiles.
+//
+// Fragment layout (measured on gfx950 with tools/mfma_probe.hip):
+//   A:   row = lane & 15,  k = (lane>>4)*8 + e   (8 CONSECUTIVE k)
+//   B:   col = lane & 15,  k = (lane>>4)*8 + e
+//   C/D: col = lane & 15,  row = (lane>>4)*4 + r
+//
+// GEMM view: M = output pixels, N = K_out, K-dim = 9*C zero-padded to a
+// multiple of 32 in the order k = (seg)*8 + ci with seg = cb*9 + tap,
+// tap = dy*3 + dx, c = cb*8 + ci. A lane's 8 A-elements are then 8
+// consecutive input channels at ONE (dy, dx) tap: a single ds_read_b128
+// from the staged row. Weights arrive PRE-PACKED in exact fragment
+// order (mine_amd/ops/conv.py): one 16-byte global load per fragment.
+//
+// Workgroup: 4 waves; output tile = one row y x 64 pixels x K_out.
+// LDS: 3 input rows x 66 pixels x C bf16, border handling (reflect for
+// the forward, zero for the transposed/data-gradient use) applied while
+// staging.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+namespace {
+
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+constexpr int kBlock = 256;
+constexpr int TILE_W = 64;          // output pixels per workgroup
+constexpr int STAGE_W = TILE_W + 2; // 66
+constexpr int MAX_C = 128;
+
+enum PadMode { PAD_REFLECT = 0, PAD_ZERO = 1 };
+
+template <int PAD>
+__device__ __forceinline__ int map_coord(int v, int n) {
+  // reflect: -1 -> 1, n -> n-2 (pad 1); zero: OOB -> -1 sentinel
+  if (PAD == PAD_REFLECT) {
+    if (v < 0) v = -v;
+    if (v >= n) v = 2 * (n - 1) - v;
+    return v;
+  }
+  return (v < 0 || v >= n) ? -1 : v;
+}
+
+template <int PAD>
+__global__ void __launch_bounds__(kBlock)
+conv3x3_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,H,W,C)
+                   const __hip_bfloat16* __restrict__ wp, // packed frags
+                   const float* __restrict__ bias,        // (K) or null
+                   __hip_bfloat16* __restrict__ out,      // (N,H,W,K)
+                   int H, int W, int C, int K) {
+  extern __shared__ __hip_bfloat16 s_in[];  // [3][STAGE_W][C]
+  const int x0 = blockIdx.x * TILE_W;
+  const int y = blockIdx.y;
+  const int n = blockIdx.z;
+  const int tid = threadIdx.x;
+
+  // ---- stage 3 reflected rows x 66 pixels x C (8-channel vectors) ----
+  const int Cv = C / 8;
+  const int total_v = 3 * STAGE_W * Cv;
+  const int64_t x_n = (int64_t)n * H * W * C;
+  for (int i = tid; i < total_v; i += kBlock) {
+    const int cv = i % Cv;
+    const int rem = i / Cv;
+    const int sx = rem % STAGE_W;      // 0..65 -> input x = x0 + sx - 1
+    const int row = rem / STAGE_W;     // 0..2  -> input y = y + row - 1
+    const int yy = map_coord<PAD>(y + row - 1, H);
+    const int xx = map_coord<PAD>(x0 + sx - 1, W);
+    bf16x8 v;
+    if (yy < 0 || xx < 0) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) v[e] = (__bf16)0.0f;
+    } else {
+      v = *reinterpret_cast<const bf16x8*>(
+          x + x_n + ((int64_t)yy * W + xx) * C + cv * 8);
+    }
+    *reinterpret_cast<bf16x8*>(s_in + (i * 8)) = v;
+  }
+  __syncthreads();
+
+  // ---- MFMA tiles ----
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int px_base = wave * 16;            // this wave's 16 pixels
+  const int i_row = lane & 15;              // A row / output pixel offset
+  const int seg_in_chunk = lane >> 4;       // 0..3
+
+  const int nseg = 9 * Cv;                  // segments of 8 k
+  const int nchunks = (nseg + 3) / 4;       // 32-wide K chunks
+  const int nK = (K + 15) / 16;             // 16-wide N chunks (K zero-padded)
+
+  f32x4 acc[4];                             // up to K=64 held at once
+  const int nK_held = nK <= 4 ? nK : 4;
+
+  for (int nc0 = 0; nc0 < nK; nc0 += nK_held) {
+#pragma unroll
+    for (int a = 0; a < 4; ++a) acc[a] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int kc = 0; kc < nchunks; ++kc) {
+      const int seg = kc * 4 + seg_in_chunk;
+      bf16x8 afrag;
+      if (seg < nseg) {
+        const int cb = seg / 9;
+        const int tap = seg - cb * 9;
+        const int dy = tap / 3, dx = tap - dy * 3;
+        // staged coords: row dy, x index (px_base + i_row) + dx
+        const int sx = px_base + i_row + dx;  // 0..65 (+2 from taps)
+        afrag = *reinterpret_cast<const bf16x8*>(
+            s_in + ((dy * STAGE_W + sx) * Cv + cb) * 8);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) afrag[e] = (__bf16)0.0f;
+      }
+#pragma unroll
+      for (int a = 0; a < 4; ++a) {
+        if (a < nK_held && nc0 + a < nK) {
+          const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+              wp + (((int64_t)(nc0 + a) * nchunks + kc) * 64 + lane) * 8);
+          acc[a] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                           acc[a], 0, 0, 0);
+        }
+      }
+    }
+
+    // ---- epilogue: bias + store ----
+    const int j = lane & 15;                 // output channel offset
+#pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      if (a < nK_held && nc0 + a < nK) {
+        const int kout = (nc0 + a) * 16 + j;
+        const float b = (bias && kout < K) ? bias[kout] : 0.0f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int pix = x0 + px_base + (lane >> 4) * 4 + r;
+          if (pix < W && kout < K) {
+            out[((int64_t)n * H + y) * W * K + (int64_t)pix * K + kout] =
+                (__hip_bfloat16)(acc[a][r] + b);
+          }
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" void mine_conv3x3_fwd(const void* x, const void* wp,
+                                 const float* bias, void* out, int N, int H,
+                                 int W, int C, int K, int pad_mode,
+                                 hipStream_t stream) {
+  const dim3 grid((W + TILE_W - 1) / TILE_W, H, N);
+  const size_t lds = 3 * STAGE_W * C * sizeof(__hip_bfloat16);
+  if (pad_mode == PAD_REFLECT)
+    hipLaunchKernelGGL(conv3x3_fwd_kernel<PAD_REFLECT>, grid, dim3(kBlock),
+                       lds, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x),
+                       reinterpret_cast<const __hip_bfloat16*>(wp), bias,
+                       reinterpret_cast<__hip_bfloat16*>(out), H, W, C, K);
+  else
+    hipLaunchKernelGGL(conv3x3_fwd_kernel<PAD_ZERO>, grid, dim3(kBlock),
+                       lds, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x),
+                       reinterpret_cast<const __hip_bfloat16*>(wp), bias,
+                       reinterpret_cast<__hip_bfloat16*>(out), H, W, C, K);
+}

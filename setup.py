@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "pad_kernels.hip"),
         os.path.join(CSRC, "bn_kernels.hip"),
         os.path.join(CSRC, "head_kernels.hip"),
+        os.path.join(CSRC, "conv_kernels.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

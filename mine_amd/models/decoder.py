@@ -42,7 +42,10 @@ class ConvBlock(nn.Module):
         self.bn = FusedBNAct(out_ch, act="elu")
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.bn(self.conv(self.pad(x)))
+        from mine_amd.ops.conv import conv3x3_reflect
+        # pad folded into the MFMA conv on the GPU fast path; the eager
+        # fallback inside conv3x3_reflect pads explicitly
+        return self.bn(conv3x3_reflect(x, self.conv.weight, self.conv.bias))
 
 
 class SplitConvBlock(nn.Module):
@@ -92,7 +95,8 @@ class SplitConvBlock(nn.Module):
             + bias_pe.view(B, S, 1, 1, K).to(y_base.dtype)
         yb = yb.view(B * S, Hb, Wb, K)
         if d:
-            y_dec = F.conv2d(self.pad(x_dec), w[:, :d], None)
+            from mine_amd.ops.conv import conv3x3_reflect
+            y_dec = conv3x3_reflect(x_dec, w[:, :d].contiguous(), None)
             yb = yb + y_dec.permute(0, 2, 3, 1)
         return self.bn(yb.permute(0, 3, 1, 2))
 
@@ -115,7 +119,8 @@ class Conv3x3Refl(nn.Module):
         self.conv = nn.Conv2d(in_ch, out_ch, 3)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.conv(self.pad(x))
+        from mine_amd.ops.conv import conv3x3_reflect
+        return conv3x3_reflect(x, self.conv.weight, self.conv.bias)
 
 
 class MPIDecoder(nn.Module):

@@ -110,5 +110,6 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
               and x.is_contiguous(memory_format=torch.channels_last))
     if usable:
         return _Conv3x3ReflFn.apply(x, w, bias)
-    return F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"),
-                    w.to(x.dtype), bias.to(x.dtype) if bias is not None else None)
+    from mine_amd.ops.pad import reflection_pad2d
+    return F.conv2d(reflection_pad2d(x, 1), w.to(x.dtype),
+                    bias.to(x.dtype) if bias is not None else None)

@@ -77,16 +77,21 @@ class _Conv3x3ReflFn(torch.autograd.Function):
                         bias.float() if bias is not None else
                         torch.empty(0, device=x.device, dtype=torch.float32),
                         out, B, H, W, C, K, 0)
-        ctx.save_for_backward(x, w)
+        # save the PADDED input for backward (one cheap HIP pad; MIOpen's
+        # tuned convolution_backward then runs exactly as in the
+        # unfused path — no recompute in the backward hot loop)
+        from mine_amd.ops.pad import reflection_pad2d
+        with torch.no_grad():
+            xp = reflection_pad2d(x.detach(), 1)
+        ctx.save_for_backward(xp, w)
+        ctx.geom = (B, C, H, W)
         ctx.has_bias = bias is not None
         return out.view(B, H, W, K).permute(0, 3, 1, 2)
 
     @staticmethod
     def backward(ctx, gy):
-        from mine_amd.ops.pad import _ReflectPadFn, reflection_pad2d
-        x, w = ctx.saved_tensors
-        with torch.no_grad():
-            xp = reflection_pad2d(x, 1)
+        xp, w = ctx.saved_tensors
+        B, C, H, W = ctx.geom
         gy = gy.contiguous(memory_format=torch.channels_last)
         gx_pad, gw, gb = torch.ops.aten.convolution_backward(
             gy, xp, w.to(xp.dtype), [w.shape[0]] if ctx.has_bias else None,
@@ -94,7 +99,6 @@ class _Conv3x3ReflFn(torch.autograd.Function):
             [True, True, ctx.has_bias])
         # fold the pad gradient back (atomic-free gather kernel)
         ext = get_extension(required=True)
-        B, C, H, W = x.shape
         flat = gx_pad.permute(0, 2, 3, 1).contiguous().reshape(-1)
         gx = ext.reflect_pad_bwd(flat, B, H, W, C, 1)
         gx = gx.view(B, H, W, C).permute(0, 3, 1, 2)

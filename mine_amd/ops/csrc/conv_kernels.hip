@@ -33,9 +33,10 @@ using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 constexpr int kBlock = 256;
-constexpr int TILE_W = 64;          // output pixels per workgroup
+constexpr int TILE_W = 64;          // output pixels per row tile
+constexpr int TILE_H = 4;           // output rows per workgroup
 constexpr int STAGE_W = TILE_W + 2; // 66
-constexpr int MAX_C = 128;
+constexpr int STAGE_H = TILE_H + 2; // 6 staged input rows (halo shared)
 
 enum PadMode { PAD_REFLECT = 0, PAD_ZERO = 1 };
 
@@ -57,22 +58,22 @@ conv3x3_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,H,W,C)
                    const float* __restrict__ bias,        // (K) or null
                    __hip_bfloat16* __restrict__ out,      // (N,H,W,K)
                    int H, int W, int C, int K) {
-  extern __shared__ __hip_bfloat16 s_in[];  // [3][STAGE_W][C]
+  extern __shared__ __hip_bfloat16 s_in[];  // [STAGE_H][STAGE_W][C]
   const int x0 = blockIdx.x * TILE_W;
-  const int y = blockIdx.y;
+  const int y0 = blockIdx.y * TILE_H;
   const int n = blockIdx.z;
   const int tid = threadIdx.x;
 
-  // ---- stage 3 reflected rows x 66 pixels x C (8-channel vectors) ----
+  // ---- stage 6 reflected rows x 66 pixels x C (8-channel vectors) ----
   const int Cv = C / 8;
-  const int total_v = 3 * STAGE_W * Cv;
+  const int total_v = STAGE_H * STAGE_W * Cv;
   const int64_t x_n = (int64_t)n * H * W * C;
   for (int i = tid; i < total_v; i += kBlock) {
     const int cv = i % Cv;
     const int rem = i / Cv;
     const int sx = rem % STAGE_W;      // 0..65 -> input x = x0 + sx - 1
-    const int row = rem / STAGE_W;     // 0..2  -> input y = y + row - 1
-    const int yy = map_coord<PAD>(y + row - 1, H);
+    const int row = rem / STAGE_W;     // 0..5  -> input y = y0 + row - 1
+    const int yy = map_coord<PAD>(y0 + row - 1, H);
     const int xx = map_coord<PAD>(x0 + sx - 1, W);
     bf16x8 v;
     if (yy < 0 || xx < 0) {
@@ -100,49 +101,53 @@ conv3x3_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,H,W,C)
   f32x4 acc[4];                             // up to K=64 held at once
   const int nK_held = nK <= 4 ? nK : 4;
 
-  for (int nc0 = 0; nc0 < nK; nc0 += nK_held) {
+  for (int row = 0; row < TILE_H; ++row) {
+    const int y = y0 + row;
+    if (y >= H) break;
+    for (int nc0 = 0; nc0 < nK; nc0 += nK_held) {
 #pragma unroll
-    for (int a = 0; a < 4; ++a) acc[a] = f32x4{0.f, 0.f, 0.f, 0.f};
+      for (int a = 0; a < 4; ++a) acc[a] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    for (int kc = 0; kc < nchunks; ++kc) {
-      const int seg = kc * 4 + seg_in_chunk;
-      bf16x8 afrag;
-      if (seg < nseg) {
-        const int cb = seg / 9;
-        const int tap = seg - cb * 9;
-        const int dy = tap / 3, dx = tap - dy * 3;
-        // staged coords: row dy, x index (px_base + i_row) + dx
-        const int sx = px_base + i_row + dx;  // 0..65 (+2 from taps)
-        afrag = *reinterpret_cast<const bf16x8*>(
-            s_in + ((dy * STAGE_W + sx) * Cv + cb) * 8);
-      } else {
+      for (int kc = 0; kc < nchunks; ++kc) {
+        const int seg = kc * 4 + seg_in_chunk;
+        bf16x8 afrag;
+        if (seg < nseg) {
+          const int cb = seg / 9;
+          const int tap = seg - cb * 9;
+          const int dy = tap / 3, dx = tap - dy * 3;
+          // staged coords: row (row + dy), x (px_base + i_row) + dx
+          const int sx = px_base + i_row + dx;  // 0..65 (+2 from taps)
+          afrag = *reinterpret_cast<const bf16x8*>(
+              s_in + (((row + dy) * STAGE_W + sx) * Cv + cb) * 8);
+        } else {
 #pragma unroll
-        for (int e = 0; e < 8; ++e) afrag[e] = (__bf16)0.0f;
+          for (int e = 0; e < 8; ++e) afrag[e] = (__bf16)0.0f;
+        }
+#pragma unroll
+        for (int a = 0; a < 4; ++a) {
+          if (a < nK_held && nc0 + a < nK) {
+            const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+                wp + (((int64_t)(nc0 + a) * nchunks + kc) * 64 + lane) * 8);
+            acc[a] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                             acc[a], 0, 0, 0);
+          }
+        }
       }
+
+      // ---- epilogue: bias + store ----
+      const int j = lane & 15;               // output channel offset
 #pragma unroll
       for (int a = 0; a < 4; ++a) {
         if (a < nK_held && nc0 + a < nK) {
-          const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-              wp + (((int64_t)(nc0 + a) * nchunks + kc) * 64 + lane) * 8);
-          acc[a] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                           acc[a], 0, 0, 0);
-        }
-      }
-    }
-
-    // ---- epilogue: bias + store ----
-    const int j = lane & 15;                 // output channel offset
+          const int kout = (nc0 + a) * 16 + j;
+          const float b = (bias && kout < K) ? bias[kout] : 0.0f;
 #pragma unroll
-    for (int a = 0; a < 4; ++a) {
-      if (a < nK_held && nc0 + a < nK) {
-        const int kout = (nc0 + a) * 16 + j;
-        const float b = (bias && kout < K) ? bias[kout] : 0.0f;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int pix = x0 + px_base + (lane >> 4) * 4 + r;
-          if (pix < W && kout < K) {
-            out[((int64_t)n * H + y) * W * K + (int64_t)pix * K + kout] =
-                (__hip_bfloat16)(acc[a][r] + b);
+          for (int r = 0; r < 4; ++r) {
+            const int pix = x0 + px_base + (lane >> 4) * 4 + r;
+            if (pix < W && kout < K) {
+              out[((int64_t)n * H + y) * W * K + (int64_t)pix * K + kout] =
+                  (__hip_bfloat16)(acc[a][r] + b);
+            }
           }
         }
       }
@@ -156,8 +161,8 @@ extern "C" void mine_conv3x3_fwd(const void* x, const void* wp,
                                  const float* bias, void* out, int N, int H,
                                  int W, int C, int K, int pad_mode,
                                  hipStream_t stream) {
-  const dim3 grid((W + TILE_W - 1) / TILE_W, H, N);
-  const size_t lds = 3 * STAGE_W * C * sizeof(__hip_bfloat16);
+  const dim3 grid((W + TILE_W - 1) / TILE_W, (H + TILE_H - 1) / TILE_H, N);
+  const size_t lds = STAGE_H * STAGE_W * C * sizeof(__hip_bfloat16);
   if (pad_mode == PAD_REFLECT)
     hipLaunchKernelGGL(conv3x3_fwd_kernel<PAD_REFLECT>, grid, dim3(kBlock),
                        lds, stream,

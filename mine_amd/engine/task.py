@@ -430,6 +430,34 @@ class SynthesisTask:
         mark("set_data")
         loss_dict, _ = self.loss_fcn(is_val=False)
         mark("forward")
+        # NaN guard (absent in the reference — SURVEY section 5c): a
+        # non-finite loss skips the update instead of poisoning the
+        # parameters. The skip decision is agreed ACROSS ranks (min over
+        # the finite flags) so every rank steps or skips together and the
+        # gradient buckets are still reduced — collectives stay matched.
+        if bool(self.config.get("training.nan_guard", True)):
+            finite = torch.isfinite(loss_dict["loss"].detach()).to(torch.int32)
+            if torch.distributed.is_initialized() and \
+                    torch.distributed.get_world_size() > 1:
+                finite = finite.clone()
+                torch.distributed.all_reduce(
+                    finite, op=torch.distributed.ReduceOp.MIN)
+            if not bool(finite.item()):
+                if self.logger:
+                    self.logger.warning(
+                        "non-finite loss at step %d; skipping update",
+                        self.global_step)
+                self._nan_skips = getattr(self, "_nan_skips", 0) + 1
+                if self.grad_engine is not None:
+                    self.grad_engine.zero_grad()
+                else:
+                    self.optimizer.zero_grad(set_to_none=False)
+                loss_dict["loss"].backward()  # keep DDP collectives matched
+                if self.grad_engine is not None:
+                    self.grad_engine.finish_step()
+                mark("backward")
+                mark("optimizer")
+                return loss_dict
         if self.grad_engine is not None:
             self.grad_engine.zero_grad()
         else:

@@ -100,3 +100,32 @@ def test_coarse_to_fine_path(tiny_config):
     d = endpoints["disparity_all_src"]
     assert (d[:, :-1] >= d[:, 1:]).all()  # sorted descending
     assert endpoints["mpi_all_src_list"][0].shape[1] == S
+
+
+def test_nan_guard_skips_update():
+    import torch
+    from mine_amd.config import default_config
+    from mine_amd.data import SyntheticMPIDataset, collate_src_tgt
+    from mine_amd.engine import SynthesisTask
+
+    cfg = default_config(**{
+        "data.name": "synthetic", "data.img_h": 64, "data.img_w": 64,
+        "mpi.num_bins_coarse": 4, "data.per_gpu_batch_size": 2,
+        "data.visible_point_count": 8, "training.amp_dtype": "fp32",
+    })
+    ds = SyntheticMPIDataset(cfg, length=2)
+    items = collate_src_tgt([ds[0], ds[1]])
+    task = SynthesisTask(cfg, device="cpu")
+
+    # poison the input -> non-finite loss
+    bad = (dict(items[0]), items[1])
+    bad[0]["img"] = items[0]["img"] * float("nan")
+    w0 = task.decoder.dispconvs["0"].conv.weight.detach().clone()
+    loss = task.train_step((bad[0], bad[1]))
+    assert not torch.isfinite(loss["loss"])
+    torch.testing.assert_close(task.decoder.dispconvs["0"].conv.weight, w0)
+    assert getattr(task, "_nan_skips", 0) == 1
+
+    # a healthy step still updates
+    task.train_step(items)
+    assert not torch.equal(task.decoder.dispconvs["0"].conv.weight, w0)

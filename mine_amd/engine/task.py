@@ -449,12 +449,12 @@ class SynthesisTask:
                         self.global_step)
                 self._nan_skips = getattr(self, "_nan_skips", 0) + 1
                 if self.grad_engine is not None:
+                    # no backward: finish_step() reduces the zeroed buckets
+                    # itself, so the collectives stay matched across ranks
                     self.grad_engine.zero_grad()
+                    self.grad_engine.finish_step()
                 else:
                     self.optimizer.zero_grad(set_to_none=False)
-                loss_dict["loss"].backward()  # keep DDP collectives matched
-                if self.grad_engine is not None:
-                    self.grad_engine.finish_step()
                 mark("backward")
                 mark("optimizer")
                 return loss_dict

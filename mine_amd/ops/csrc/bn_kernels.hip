@@ -232,6 +232,146 @@ inline int grid_elems(int64_t total) {
   return (int)(g < 65535 ? g : 65535);
 }
 
+
+// ---------------------------------------------------------------------------
+// vectorized variants: V consecutive channels per thread (one 16-byte
+// load per element row) — the decoder's C=16 bf16 layers go from 2-byte
+// to 16-byte access granularity.
+// ---------------------------------------------------------------------------
+
+template <typename T, int V>
+struct alignas(sizeof(T) * V) BnVec {
+  T v[V];
+};
+
+inline __host__ __device__ int chan_group_v(int Cv) {
+  int g = 1;
+  while (g < Cv && g < 64) g <<= 1;
+  return g;
+}
+
+template <typename T, int V>
+__global__ void __launch_bounds__(kBlock)
+bn_stats_vec_kernel(const T* __restrict__ x, float* __restrict__ sums,
+                    int64_t M, int C, int cgv) {
+  using Vec = BnVec<T, V>;
+  const int Cv = C / V;
+  const int c0v = blockIdx.y * cgv;
+  const int ncv = min(cgv, Cv - c0v);
+  __shared__ float s_sum[512], s_sq[512];
+  for (int i = threadIdx.x; i < cgv * V; i += kBlock) {
+    s_sum[i] = 0.0f;
+    s_sq[i] = 0.0f;
+  }
+  __syncthreads();
+
+  const int lane_cv = threadIdx.x & (cgv - 1);
+  const int rows_per_blk = kBlock / cgv;
+  const int row0 = blockIdx.x * rows_per_blk + threadIdx.x / cgv;
+  const int rstride = gridDim.x * rows_per_blk;
+  if (lane_cv < ncv) {
+    float acc[V], asq[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) acc[j] = asq[j] = 0.0f;
+    const Vec* xv = reinterpret_cast<const Vec*>(x);
+    for (int64_t m = row0; m < M; m += rstride) {
+      const Vec val = xv[m * Cv + c0v + lane_cv];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float f = (float)val.v[j];
+        acc[j] += f;
+        asq[j] += f * f;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      atomicAdd(&s_sum[lane_cv * V + j], acc[j]);
+      atomicAdd(&s_sq[lane_cv * V + j], asq[j]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < ncv * V; i += kBlock) {
+    atomicAdd(&sums[c0v * V + i], s_sum[i]);
+    atomicAdd(&sums[C + c0v * V + i], s_sq[i]);
+  }
+}
+
+template <typename T, int V>
+__global__ void __launch_bounds__(kBlock)
+bn_act_bwd_reduce_vec_kernel(const T* __restrict__ x,
+                             const T* __restrict__ res,
+                             const T* __restrict__ gy,
+                             const float* __restrict__ mean,
+                             const float* __restrict__ invstd,
+                             const float* __restrict__ gamma,
+                             const float* __restrict__ beta,
+                             float* __restrict__ out,  // (2,C)
+                             int64_t M, int C, int act, int cgv) {
+  using Vec = BnVec<T, V>;
+  const int Cv = C / V;
+  const int c0v = blockIdx.y * cgv;
+  const int ncv = min(cgv, Cv - c0v);
+  __shared__ float s_db[512], s_dg[512];
+  for (int i = threadIdx.x; i < cgv * V; i += kBlock) {
+    s_db[i] = 0.0f;
+    s_dg[i] = 0.0f;
+  }
+  __syncthreads();
+  const int lane_cv = threadIdx.x & (cgv - 1);
+  const int rows_per_blk = kBlock / cgv;
+  const int row0 = blockIdx.x * rows_per_blk + threadIdx.x / cgv;
+  const int rstride = gridDim.x * rows_per_blk;
+  if (lane_cv < ncv) {
+    const int cb = (c0v + lane_cv) * V;
+    float mu[V], is[V], ga[V], be[V], db[V], dg[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      mu[j] = mean[cb + j];
+      is[j] = invstd[cb + j];
+      ga[j] = gamma[cb + j];
+      be[j] = beta[cb + j];
+      db[j] = dg[j] = 0.0f;
+    }
+    const Vec* xv = reinterpret_cast<const Vec*>(x);
+    const Vec* gv = reinterpret_cast<const Vec*>(gy);
+    const Vec* rv = reinterpret_cast<const Vec*>(res);
+    for (int64_t m = row0; m < M; m += rstride) {
+      const int64_t off = m * Cv + c0v + lane_cv;
+      const Vec xval = xv[off];
+      const Vec gval = gv[off];
+      Vec rval;
+      if (act == ACT_ADD_RELU) rval = rv[off];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float xh = ((float)xval.v[j] - mu[j]) * is[j];
+        float z = xh * ga[j] + be[j];
+        if (act == ACT_ADD_RELU) z += (float)rval.v[j];
+        const float g = (float)gval.v[j] * act_grad(act, z);
+        db[j] += g;
+        dg[j] += g * xh;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      atomicAdd(&s_db[lane_cv * V + j], db[j]);
+      atomicAdd(&s_dg[lane_cv * V + j], dg[j]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < ncv * V; i += kBlock) {
+    atomicAdd(&out[c0v * V + i], s_db[i]);
+    atomicAdd(&out[C + c0v * V + i], s_dg[i]);
+  }
+}
+
+inline dim3 grid_reduce_v(int64_t M, int Cv, int cgv) {
+  const int rows_per_blk = kBlock / cgv;
+  int64_t rows = (M + rows_per_blk - 1) / rows_per_blk;
+  int gx = (int)(rows < 2048 ? rows : 2048);
+  if (gx < 1) gx = 1;
+  return dim3(gx, (Cv + cgv - 1) / cgv);
+}
+
 inline dim3 grid_reduce(int64_t M, int C, int cg) {
   // enough slabs to fill 256 CUs x a few blocks, bounded
   const int rows_per_blk = kBlock / cg;
@@ -246,6 +386,15 @@ inline dim3 grid_reduce(int64_t M, int C, int cg) {
 #define EXPORT_BN(SUF, T)                                                      \
   extern "C" void mine_bn_stats_##SUF(const void* x, float* sums, int64_t M,   \
                                       int C, hipStream_t s) {                  \
+    const int V = 16 / (int)sizeof(T);                                         \
+    if (C % V == 0) {                                                          \
+      const int Cv = C / V;                                                    \
+      const int cgv = chan_group_v(Cv);                                        \
+      hipLaunchKernelGGL((bn_stats_vec_kernel<T, 16 / (int)sizeof(T)>),        \
+                         grid_reduce_v(M, Cv, cgv), dim3(kBlock), 0, s,        \
+                         reinterpret_cast<const T*>(x), sums, M, C, cgv);      \
+      return;                                                                  \
+    }                                                                          \
     const int cg = chan_group(C);                                              \
     hipLaunchKernelGGL(bn_stats_kernel<T>, grid_reduce(M, C, cg),              \
                        dim3(kBlock), 0, s, reinterpret_cast<const T*>(x),      \
@@ -264,6 +413,18 @@ inline dim3 grid_reduce(int64_t M, int C, int cg) {
       const void* x, const void* res, const void* gy, const float* mean,       \
       const float* invstd, const float* gamma, const float* beta, float* out,  \
       int64_t M, int C, int act, hipStream_t s) {                              \
+    const int V = 16 / (int)sizeof(T);                                         \
+    if (C % V == 0) {                                                          \
+      const int Cv = C / V;                                                    \
+      const int cgv = chan_group_v(Cv);                                        \
+      hipLaunchKernelGGL(                                                      \
+          (bn_act_bwd_reduce_vec_kernel<T, 16 / (int)sizeof(T)>),              \
+          grid_reduce_v(M, Cv, cgv), dim3(kBlock), 0, s,                       \
+          reinterpret_cast<const T*>(x), reinterpret_cast<const T*>(res),      \
+          reinterpret_cast<const T*>(gy), mean, invstd, gamma, beta, out, M,   \
+          C, act, cgv);                                                        \
+      return;                                                                  \
+    }                                                                          \
     const int cg = chan_group(C);                                              \
     hipLaunchKernelGGL(bn_act_bwd_reduce_kernel<T>, grid_reduce(M, C, cg),     \
                        dim3(kBlock), 0, s, reinterpret_cast<const T*>(x),      \

@@ -113,6 +113,17 @@ src_composite_fwd_kernel(const float* __restrict__ mpi,
       Wsum += w;
       Nsum += w * d;  // src-view z of plane s is its depth
       A *= (t + 1e-6f);
+      if (A < 1e-14f) {  // dead transmittance: remaining planes add ~0
+        if (BLEND) {
+          // the blended tail is c = A*I + (1-A)*rgb ~= rgb: copy through
+          for (int s2 = s + 1; s2 < S; ++s2) {
+            const int64_t o = mpi_b + ((int64_t)s2 * HW + pix) * 4;
+            *reinterpret_cast<float4*>(mpi_blend + o) =
+                *reinterpret_cast<const float4*>(mpi + o);
+          }
+        }
+        break;
+      }
     }
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
                            : (Nsum / (Wsum + 1e-5f));
@@ -249,6 +260,7 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
           px, (float)Ad, I, gR, gD, d, delta, D, Wp, gC);
       TotalP += (double)pt.dA * Ad;
       Ad *= (double)(pt.t + 1e-6f);
+      if (Ad < 1e-14) break;
     }
 
     // ---- pass 3: emit gradients ----
@@ -283,6 +295,19 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
       }
       *reinterpret_cast<float4*>(grad_mpi + mpi_b + ((int64_t)s * HW + pix) * 4) = g;
       Ad *= (double)u;
+      if (Ad < 1e-14) {
+        // tail: w ~ 0, A ~ 0 -> drgb = dc ~= gC, dsigma = gCs (blend
+        // pass-through); zeros without blend. grad_mpi is at::empty.
+        for (int s2 = s + 1; s2 < S; ++s2) {
+          const int64_t o = mpi_b + ((int64_t)s2 * HW + pix) * 4;
+          float4 gt = make_float4(0.f, 0.f, 0.f, 0.f);
+          if (BLEND && g_blend) {
+            gt = *reinterpret_cast<const float4*>(g_blend + o);
+          }
+          *reinterpret_cast<float4*>(grad_mpi + o) = gt;
+        }
+        break;
+      }
     }
   }
 }

@@ -260,3 +260,46 @@ def test_colmap_database(tmp_path):
     assert pair_id_to_image_ids(pid) == (min(i1, i2), max(i1, i2))
     assert db.execute("SELECT pair_id FROM matches").fetchone()[0] == pid
     db.close()
+
+
+def test_eval_pairs_protocol(tmp_path):
+    """tools/eval_pairs.py consumes the reference's RealEstate10K
+    validation-pair JSONL schema end-to-end on synthetic frames."""
+    import json
+    import subprocess
+    import sys as _sys
+    from PIL import Image as PILImage
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    rng = np.random.default_rng(5)
+    seq = "seq0"
+    os.makedirs(tmp_path / "frames" / seq)
+    for ts in ("100", "105", "110"):
+        arr = (rng.uniform(0, 1, (48, 64, 3)) * 255).astype(np.uint8)
+        PILImage.fromarray(arr).save(tmp_path / "frames" / seq / f"{ts}.png")
+
+    def obj(ts, tx):
+        return {"sequence_id": seq,
+                "camera_intrinsics": [0.5, 0.6, 0.5, 0.5],
+                "camera_pose": [1, 0, 0, tx, 0, 1, 0, 0, 0, 0, 1, 0],
+                "frame_ts": ts}
+
+    entry = {"sequence_id": seq, "src_img_obj": obj("100", 0.0),
+             "tgt_img_obj_5_frames": obj("105", 0.05),
+             "tgt_img_obj_10_frames": obj("110", 0.1)}
+    pairs = tmp_path / "pairs.json"
+    pairs.write_text(json.dumps(entry) + "\n")
+
+    extra = {"data.name": "realestate10k", "data.img_h": 32, "data.img_w": 48,
+             "mpi.num_bins_coarse": 4, "data.visible_point_count": 8,
+             "training.amp_dtype": "fp32"}
+    r = subprocess.run(
+        [_sys.executable, os.path.join(root, "tools", "eval_pairs.py"),
+         "--pairs", str(pairs), "--data_root", str(tmp_path / "frames"),
+         "--extra_config", json.dumps(extra)],
+        cwd=root, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["5_frames"]["n_pairs"] == 1
+    assert out["10_frames"]["n_pairs"] == 1
+    assert out["5_frames"]["psnr"] > 0

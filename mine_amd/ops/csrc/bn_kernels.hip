@@ -364,6 +364,86 @@ bn_act_bwd_reduce_vec_kernel(const T* __restrict__ x,
   }
 }
 
+
+template <typename T, int V>
+__global__ void __launch_bounds__(kBlock)
+bn_act_fwd_vec_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                      const float* __restrict__ mean,
+                      const float* __restrict__ invstd,
+                      const float* __restrict__ gamma,
+                      const float* __restrict__ beta, T* __restrict__ y,
+                      int64_t M, int C, int act) {
+  using Vec = BnVec<T, V>;
+  const int Cv = C / V;
+  const int64_t total = M * Cv;
+  const Vec* xv = reinterpret_cast<const Vec*>(x);
+  const Vec* rv = reinterpret_cast<const Vec*>(res);
+  Vec* yv = reinterpret_cast<Vec*>(y);
+  for (int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * kBlock) {
+    const int cb = (int)(i % Cv) * V;
+    const Vec xval = xv[i];
+    Vec rval;
+    if (act == ACT_ADD_RELU) rval = rv[i];
+    Vec out;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      float z = ((float)xval.v[j] - mean[cb + j]) * invstd[cb + j] *
+                    gamma[cb + j] + beta[cb + j];
+      if (act == ACT_ADD_RELU) {
+        z += (float)rval.v[j];
+        out.v[j] = (T)(z > 0.0f ? z : 0.0f);
+      } else {
+        out.v[j] = (T)act_fwd(act, z);
+      }
+    }
+    yv[i] = out;
+  }
+}
+
+template <typename T, int V>
+__global__ void __launch_bounds__(kBlock)
+bn_act_bwd_dx_vec_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                         const T* __restrict__ gy,
+                         const float* __restrict__ mean,
+                         const float* __restrict__ invstd,
+                         const float* __restrict__ gamma,
+                         const float* __restrict__ beta,
+                         const float* __restrict__ red,  // (2,C)
+                         T* __restrict__ dx, T* __restrict__ dres,
+                         int64_t M, int C, int act) {
+  using Vec = BnVec<T, V>;
+  const int Cv = C / V;
+  const float invM = 1.0f / (float)M;
+  const int64_t total = M * Cv;
+  const Vec* xv = reinterpret_cast<const Vec*>(x);
+  const Vec* rv = reinterpret_cast<const Vec*>(res);
+  const Vec* gv = reinterpret_cast<const Vec*>(gy);
+  Vec* dxv = reinterpret_cast<Vec*>(dx);
+  Vec* drv = reinterpret_cast<Vec*>(dres);
+  for (int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * kBlock) {
+    const int cb = (int)(i % Cv) * V;
+    const Vec xval = xv[i];
+    const Vec gval = gv[i];
+    Vec rval, dxo, dro;
+    if (act == ACT_ADD_RELU) rval = rv[i];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const int c = cb + j;
+      const float xh = ((float)xval.v[j] - mean[c]) * invstd[c];
+      float z = xh * gamma[c] + beta[c];
+      if (act == ACT_ADD_RELU) z += (float)rval.v[j];
+      const float g = (float)gval.v[j] * act_grad(act, z);
+      if (act == ACT_ADD_RELU) dro.v[j] = (T)g;
+      dxo.v[j] = (T)(gamma[c] * invstd[c] *
+                     (g - (red[c] + xh * red[C + c]) * invM));
+    }
+    dxv[i] = dxo;
+    if (act == ACT_ADD_RELU) drv[i] = dro;
+  }
+}
+
 inline dim3 grid_reduce_v(int64_t M, int Cv, int cgv) {
   const int rows_per_blk = kBlock / cgv;
   int64_t rows = (M + rows_per_blk - 1) / rows_per_blk;
@@ -404,6 +484,15 @@ inline dim3 grid_reduce(int64_t M, int C, int cg) {
       const void* x, const void* res, const float* mean, const float* invstd,  \
       const float* gamma, const float* beta, void* y, int64_t M, int C,        \
       int act, hipStream_t s) {                                                \
+    const int V = 16 / (int)sizeof(T);                                         \
+    if (C % V == 0) {                                                          \
+      hipLaunchKernelGGL((bn_act_fwd_vec_kernel<T, 16 / (int)sizeof(T)>),      \
+                         dim3(grid_elems(M * (C / V))), dim3(kBlock), 0, s,    \
+                         reinterpret_cast<const T*>(x),                        \
+                         reinterpret_cast<const T*>(res), mean, invstd,        \
+                         gamma, beta, reinterpret_cast<T*>(y), M, C, act);     \
+      return;                                                                  \
+    }                                                                          \
     hipLaunchKernelGGL(bn_act_fwd_kernel<T>, dim3(grid_elems(M * C)),          \
                        dim3(kBlock), 0, s, reinterpret_cast<const T*>(x),      \
                        reinterpret_cast<const T*>(res), mean, invstd, gamma,   \
@@ -437,6 +526,17 @@ inline dim3 grid_reduce(int64_t M, int C, int cg) {
       const float* invstd, const float* gamma, const float* beta,              \
       const float* red, void* dx, void* dres, int64_t M, int C, int act,       \
       hipStream_t s) {                                                         \
+    const int V = 16 / (int)sizeof(T);                                         \
+    if (C % V == 0) {                                                          \
+      hipLaunchKernelGGL((bn_act_bwd_dx_vec_kernel<T, 16 / (int)sizeof(T)>),   \
+                         dim3(grid_elems(M * (C / V))), dim3(kBlock), 0, s,    \
+                         reinterpret_cast<const T*>(x),                        \
+                         reinterpret_cast<const T*>(res),                      \
+                         reinterpret_cast<const T*>(gy), mean, invstd, gamma,  \
+                         beta, red, reinterpret_cast<T*>(dx),                  \
+                         reinterpret_cast<T*>(dres), M, C, act);               \
+      return;                                                                  \
+    }                                                                          \
     hipLaunchKernelGGL(bn_act_bwd_dx_kernel<T>, dim3(grid_elems(M * C)),       \
                        dim3(kBlock), 0, s, reinterpret_cast<const T*>(x),      \
                        reinterpret_cast<const T*>(res),                        \

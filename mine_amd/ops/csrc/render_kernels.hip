@@ -360,7 +360,8 @@ struct PlaneSample {
   float inb;
 };
 
-DEV PlaneSample sample_plane(const float* __restrict__ mpi_b, int s, int HW,
+__device__ __attribute__((noinline)) PlaneSample
+sample_plane(const float* __restrict__ mpi_b, int s, int HW,
                              const float* __restrict__ s_geom, float d,
                              const float3x3& M, float3 tv,
                              int x, int y, int W, int H, TapRef* tap_out) {
@@ -488,16 +489,35 @@ DEV void scatter4(float* base, const TapRef& t, int W, float4 g) {
 #undef SC
 }
 
-// Per-plane (t, e) for the tgt backward — noinline for the same exact-
-// cancellation reason as src_plane_term above.
-template <bool BG_INF>
-__device__ __attribute__((noinline)) float2
-tgt_plane_term(float4 rgbs, float vz, float delta, float3 gR, float gD,
-               float D, float Wp) {
-  const float t = __expf(-rgbs.w * delta);
-  const float e = rgbs.x * gR.x + rgbs.y * gR.y + rgbs.z * gR.z +
-                  gD * (BG_INF ? (vz - 1000.0f) : (vz - D) / Wp);
-  return make_float2(t, e);
+// Per-plane inter-plane distance — noinline so both backward passes get
+// bitwise-identical values (exact-cancellation requirement, see
+// src_plane_term).
+__device__ __attribute__((noinline)) float plane_delta(float3 a, float3 b) {
+  const float dx = b.x - a.x, dy = b.y - a.y, dz = b.z - a.z;
+  return sqrtf(dx * dx + dy * dy + dz * dz);
+}
+
+// Per-plane backward components for the tgt kernel. e depends on the
+// composite totals (D, Wsum) which are only known after a full pass, but
+// it is LINEAR in them: e = r + qv*vz + qc with pass-independent
+// r = rgb.gR. The suffix sums therefore split into three component sums
+// (cr, cv, cw) accumulated WITHOUT D, and the /u-amplified telescoping
+// stays exact per component because this one noinline instance feeds
+// both passes bitwise-identical terms.
+struct TgtTerm {
+  float t, r, cr, cv, cw;
+};
+
+__device__ __attribute__((noinline)) TgtTerm
+tgt_plane_terms(float4 rgbs, float vz, float delta, float3 gR) {
+  TgtTerm o;
+  o.t = __expf(-rgbs.w * delta);
+  o.r = rgbs.x * gR.x + rgbs.y * gR.y + rgbs.z * gR.z;
+  const float omt = 1.0f - o.t;
+  o.cr = omt * o.r;
+  o.cv = omt * vz;
+  o.cw = omt;
+  return o;
 }
 
 template <bool BG_INF>
@@ -539,8 +559,10 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
     }
     const float gD = g_depth ? g_depth[(int64_t)b * HW + pix] : 0.0f;
 
-    // ---- pass 1 ----
+    // ---- pass A: composite totals + D-independent suffix components ----
     float A = 1.0f, Wsum = 0.0f, Nsum = 0.0f;
+    double TaR = 0.0, TaV = 0.0, TaW = 0.0;
+    double Ad = 1.0;
     PlaneSample cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv,
                                    x, y, W, H, nullptr);
     for (int s = 0; s < S; ++s) {
@@ -549,51 +571,30 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       if (s + 1 < S) {
         nxt = sample_plane(mpi_b, s + 1, HW, s_geom, s_depth[s + 1], M, tv,
                            x, y, W, H, nullptr);
-        const float dx = nxt.v.x - cur.v.x, dy = nxt.v.y - cur.v.y,
-                    dz = nxt.v.z - cur.v.z;
-        delta = sqrtf(dx * dx + dy * dy + dz * dz);
+        delta = plane_delta(cur.v, nxt.v);
       } else {
         delta = 1e3f;
       }
-      const float t = __expf(-cur.rgbs.w * delta);
-      const float w = A * (1.0f - t);
+      const TgtTerm tt = tgt_plane_terms(cur.rgbs, cur.v.z, delta, gR);
+      const float w = A * (1.0f - tt.t);
       Wsum += w;
       Nsum += w * cur.v.z;
-      A *= (t + 1e-6f);
+      TaR += (double)tt.cr * Ad;
+      TaV += (double)tt.cv * Ad;
+      TaW += (double)tt.cw * Ad;
+      A *= (tt.t + 1e-6f);
+      Ad *= (double)(tt.t + 1e-6f);
       cur = nxt;
-      if (A < 1e-14f) break;  // dead transmittance: tail adds ~0
+      if (Ad < 1e-14) break;  // dead transmittance: tail adds ~0
     }
     const float Wp = Wsum + 1e-5f;
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
+    // e = r + qv*vz + qc
+    const float qv = BG_INF ? gD : gD / Wp;
+    const float qc = BG_INF ? (-1000.0f * gD) : (-gD / Wp * D);
 
-    // ---- pass 2 ---- (fp64 transmittance + suffix accumulators;
-    // see src bwd note)
-    double TotalP = 0.0;
-    double Ad = 1.0;
-    cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
-                       nullptr);
-    for (int s = 0; s < S; ++s) {
-      float delta;
-      PlaneSample nxt;
-      if (s + 1 < S) {
-        nxt = sample_plane(mpi_b, s + 1, HW, s_geom, s_depth[s + 1], M, tv,
-                           x, y, W, H, nullptr);
-        const float dx = nxt.v.x - cur.v.x, dy = nxt.v.y - cur.v.y,
-                    dz = nxt.v.z - cur.v.z;
-        delta = sqrtf(dx * dx + dy * dy + dz * dz);
-      } else {
-        delta = 1e3f;
-      }
-      const float2 te = tgt_plane_term<BG_INF>(cur.rgbs, cur.v.z, delta,
-                                               gR, gD, D, Wp);
-      TotalP += (double)((1.0f - te.x) * te.y) * Ad;
-      Ad *= (double)(te.x + 1e-6f);
-      cur = nxt;
-      if (Ad < 1e-14) break;
-    }
-
-    // ---- pass 3: emit gradients, bilinear scatter ----
-    double prefix = 0.0;
+    // ---- pass B: emit gradients, bilinear scatter ----
+    double prR = 0.0, prV = 0.0, prW = 0.0;
     Ad = 1.0;
     TapRef tap;
     cur = sample_plane(mpi_b, 0, HW, s_geom, s_depth[0], M, tv, x, y, W, H,
@@ -605,20 +606,24 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       if (s + 1 < S) {
         nxt = sample_plane(mpi_b, s + 1, HW, s_geom, s_depth[s + 1], M, tv,
                            x, y, W, H, &ntap);
-        const float dx = nxt.v.x - cur.v.x, dy = nxt.v.y - cur.v.y,
-                    dz = nxt.v.z - cur.v.z;
-        delta = sqrtf(dx * dx + dy * dy + dz * dz);
+        delta = plane_delta(cur.v, nxt.v);
       } else {
         delta = 1e3f;
       }
-      const float2 te = tgt_plane_term<BG_INF>(cur.rgbs, cur.v.z, delta,
-                                               gR, gD, D, Wp);
-      const float t = te.x;
+      const TgtTerm tt = tgt_plane_terms(cur.rgbs, cur.v.z, delta, gR);
+      const float t = tt.t;
       const float u = t + 1e-6f;
       const float Af = (float)Ad;
       const float w = Af * (1.0f - t);
-      prefix += (double)((1.0f - t) * te.y) * Ad;
-      const float dt = -Af * te.y + (float)((TotalP - prefix) / (double)u);
+      prR += (double)tt.cr * Ad;
+      prV += (double)tt.cv * Ad;
+      prW += (double)tt.cw * Ad;
+      // per-component exact telescoping, recombined with the (qv, qc)
+      // coefficients in fp64
+      const double suffix = (TaR - prR) + (double)qv * (TaV - prV) +
+                            (double)qc * (TaW - prW);
+      const float e = tt.r + qv * cur.v.z + qc;
+      const float dt = -Af * e + (float)(suffix / (double)u);
       // culled sigma contributed nothing -> no gradient through it
       const float dsigma = (cur.v.z < 0.0f) ? 0.0f : dt * (-delta * t);
       scatter4(gm_b + (int64_t)s * HW * 4, tap, W,

@@ -32,7 +32,7 @@ def main() -> int:
     p.add_argument("--planes", type=int, default=64)
     p.add_argument("--height", type=int, default=256)
     p.add_argument("--width", type=int, default=384)
-    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp16", "fp32"])
     p.add_argument("--dataset", type=str, default="realestate10k")
     p.add_argument("--timers", action="store_true",
                    help="print a sync-bracketed per-phase breakdown (diagnostic "

@@ -107,8 +107,13 @@ def main() -> int:
                   for k, v in task.pop_phase_times().items()}
         print(json.dumps({"phase_ms": phases}), file=sys.stderr)
     if rank == 0:
+        metric = "train imgs/sec (whole node) RealEstate10K 384x256 N=64"
+        if (args.dataset, args.height, args.width, args.planes) != \
+                ("realestate10k", 256, 384, 64):
+            metric = (f"train imgs/sec (whole node) {args.dataset} "
+                      f"{args.width}x{args.height} N={args.planes}")
         result = {
-            "metric": "train imgs/sec (whole node) RealEstate10K 384x256 N=64",
+            "metric": metric,
             "value": round(imgs_per_sec, 3),
             "unit": "imgs/sec",
             "n_gpus": world_size,

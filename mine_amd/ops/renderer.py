@@ -126,11 +126,9 @@ def render_src_view(mpi: torch.Tensor,
     grid = tr.make_meshgrid(mpi.shape[2], mpi.shape[3], device=mpi.device)
     xyz = tr.src_plane_xyz(grid, disparity, K_inv)
     if use_alpha:
-        rgb_syn, weights = tr.alpha_composite(sigma, rgb)
+        # no RGB blending under alpha compositing (ref mpi_rendering.py:19)
+        rgb_syn, _ = tr.alpha_composite(sigma, rgb)
         depth_syn, _ = tr.alpha_composite(sigma, xyz[:, :, 2:])
-        acc = torch.zeros_like(rgb)  # no blending under alpha (ref mpi_rendering.py:19)
-        if blend:
-            return rgb_syn, depth_syn, mpi
         return rgb_syn, depth_syn, mpi
     rgb_syn, depth_syn, acc, weights = tr.volume_composite(rgb, sigma, xyz, bg_depth_inf)
     if blend:

@@ -19,8 +19,8 @@ from __future__ import annotations
 
 import os
 import struct
-from dataclasses import dataclass, field
-from typing import Dict, List, Tuple
+from dataclasses import dataclass
+from typing import Dict, Tuple
 
 import numpy as np
 

@@ -26,7 +26,7 @@ from __future__ import annotations
 
 import os
 import random
-from typing import Dict, List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 import torch

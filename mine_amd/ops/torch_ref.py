@@ -18,7 +18,7 @@ cliffs exactly (ref operations/mpi_rendering.py):
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn.functional as F

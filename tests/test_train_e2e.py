@@ -49,3 +49,24 @@ def test_video_cli_writes_frames(tmp_path):
     assert r.returncode == 0, r.stderr[-3000:]
     frames = list((tmp_path / "vid" / "frames").glob("*.png"))
     assert len(frames) == 3
+
+
+def test_bench_distributed_torchrun_cpu():
+    """The driver's exact bench invocation shape (torchrun, one rank per
+    GPU) — exercised with gloo on CPU, world_size 2."""
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29611", os.path.join(ROOT, "bench.py"),
+         "--gpus", "2", "--steps", "1", "--warmup", "1", "--batch", "1",
+         "--height", "128", "--width", "192", "--planes", "4",
+         "--dtype", "fp32"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=900)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["value"] > 0

@@ -1,0 +1,123 @@
+"""Property-based invariants of the rendering math (hypothesis)."""
+import os
+import sys
+
+import numpy as np
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
+from mine_amd.ops import torch_ref as tr
+from mine_amd.utils.geometry import inverse_3x3, inverse_rigid_4x4
+
+_SETTINGS = dict(max_examples=25, deadline=None)
+
+
+@given(seed=st.integers(0, 10**6), s=st.integers(2, 12),
+       sigma_scale=st.floats(0.01, 20.0))
+@settings(**_SETTINGS)
+def test_volume_composite_weight_invariants(seed, s, sigma_scale):
+    g = torch.Generator().manual_seed(seed)
+    B, H, W = 1, 3, 4
+    rgb = torch.rand(B, s, 3, H, W, generator=g)
+    sigma = torch.rand(B, s, 1, H, W, generator=g) * sigma_scale
+    disp, _ = torch.sort(torch.rand(B, s, generator=g) * 0.9 + 0.05,
+                         descending=True, dim=1)
+    K_inv = torch.eye(3).unsqueeze(0)
+    xyz = tr.src_plane_xyz(tr.make_meshgrid(H, W), disp, K_inv)
+    rgb_out, depth_out, acc, weights = tr.volume_composite(
+        rgb, sigma, xyz, bg_depth_inf=False)
+
+    assert (weights >= 0).all()
+    # the +1e-6 cumprod bias can push the sum marginally over 1
+    assert (weights.sum(1) <= 1.0 + 1e-3 * s).all()
+    # composited rgb is a sub-convex combination of the plane colors
+    assert (rgb_out <= rgb.amax(dim=1) + 1e-5).all()
+    assert (rgb_out >= 0).all()
+    assert torch.isfinite(depth_out).all()
+
+
+@given(seed=st.integers(0, 10**6), n_samples=st.integers(1, 32))
+@settings(**_SETTINGS)
+def test_sample_pdf_within_bounds(seed, n_samples):
+    g = torch.Generator().manual_seed(seed)
+    vals, _ = torch.sort(torch.rand(2, 1, 1, 8, generator=g), dim=-1,
+                         descending=True)
+    w = torch.rand(2, 1, 1, 8, generator=g)
+    out = tr.sample_pdf(vals, w, n_samples)
+    assert out.shape == (2, 1, 1, n_samples)
+    lo = vals.amin(dim=-1, keepdim=True) - 1e-6
+    hi = vals.amax(dim=-1, keepdim=True) + 1e-6
+    assert (out >= lo).all() and (out <= hi).all()
+
+
+@given(seed=st.integers(0, 10**6))
+@settings(**_SETTINGS)
+def test_reflect_pad_crop_identity(seed):
+    from mine_amd.ops.pad import reflection_pad2d
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(2, 3, 5, 7, generator=g)
+    y = reflection_pad2d(x, 1)
+    torch.testing.assert_close(y[:, :, 1:-1, 1:-1], x)
+    # border mirrors: y[., 0] == x[., 1]
+    torch.testing.assert_close(y[:, :, 0, 1:-1], x[:, :, 1, :])
+    torch.testing.assert_close(y[:, :, 1:-1, 0], x[:, :, :, 1])
+
+
+@given(seed=st.integers(0, 10**6))
+@settings(**_SETTINGS)
+def test_rigid_inverse_roundtrip(seed):
+    g = torch.Generator().manual_seed(seed)
+    aa = torch.randn(3, generator=g)
+    th = aa.norm()
+    k = aa / (th + 1e-9)
+    K = torch.tensor([[0, -k[2], k[1]], [k[2], 0, -k[0]], [-k[1], k[0], 0.0]])
+    R = torch.eye(3) + th.sin() * K + (1 - th.cos()) * (K @ K)
+    G = torch.eye(4)
+    G[:3, :3] = R
+    G[:3, 3] = torch.randn(3, generator=g)
+    Gi = inverse_rigid_4x4(G.unsqueeze(0))[0]
+    torch.testing.assert_close(G @ Gi, torch.eye(4), rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(Gi, torch.inverse(G), rtol=1e-4, atol=1e-5)
+
+
+@given(seed=st.integers(0, 10**6), d=st.floats(0.5, 100.0))
+@settings(**_SETTINGS)
+def test_homography_inverse_consistency(seed, d):
+    """H_src_tgt from the closed-form inverse really inverts the forward
+    plane homography at arbitrary poses/depths."""
+    g = torch.Generator().manual_seed(seed)
+    aa = 0.2 * torch.randn(3, generator=g)
+    th = aa.norm()
+    k = aa / (th + 1e-9)
+    Kx = torch.tensor([[0, -k[2], k[1]], [k[2], 0, -k[0]], [-k[1], k[0], 0.0]])
+    R = torch.eye(3) + th.sin() * Kx + (1 - th.cos()) * (Kx @ Kx)
+    t = 0.3 * torch.randn(3, generator=g)
+    G = torch.eye(4)
+    G[:3, :3] = R
+    G[:3, 3] = t
+    K = torch.tensor([[40.0, 0, 16], [0, 40.0, 12], [0, 0, 1]])
+    K_inv = torch.inverse(K)
+    depths = torch.tensor([[d]], dtype=torch.float32)
+
+    H_inv = tr.homography_tgt_to_src(G.unsqueeze(0), depths,
+                                     K_inv.unsqueeze(0), K.unsqueeze(0))[0, 0]
+    tn = torch.zeros(3, 3)
+    tn[:, 2] = t
+    H_fwd = K @ (R + tn / d) @ K_inv
+    prod = H_fwd @ H_inv
+    prod = prod / prod[2, 2]
+    torch.testing.assert_close(prod, torch.eye(3), rtol=1e-3, atol=1e-3)
+
+
+@given(seed=st.integers(0, 10**6))
+@settings(**_SETTINGS)
+def test_inverse_3x3_random_wellconditioned(seed):
+    g = torch.Generator().manual_seed(seed)
+    A = torch.randn(4, 3, 3, generator=g) + 3.0 * torch.eye(3)
+    Ai = inverse_3x3(A)
+    torch.testing.assert_close(torch.matmul(A, Ai),
+                               torch.eye(3).expand(4, 3, 3),
+                               rtol=1e-3, atol=1e-3)

@@ -143,9 +143,12 @@ class SynthesisTask:
                 for m in list(self.backbone.modules()) + list(self.decoder.modules()):
                     if isinstance(m, FusedBNAct):
                         m.sync = True
+            ar_dtype = {"fp32": None, "bf16": torch.bfloat16}[
+                str(config.get("training.grad_allreduce_dtype", "fp32"))]
             self.grad_engine = GradAllReduceEngine(
                 [self.backbone, self.decoder],
-                bucket_mb=float(config.get("training.grad_bucket_mb", 25)))
+                bucket_mb=float(config.get("training.grad_bucket_mb", 25)),
+                allreduce_dtype=ar_dtype)
             self.lr_scheduler = torch.optim.lr_scheduler.MultiStepLR(
                 self.optimizer, config["lr.decay_steps"],
                 gamma=config["lr.decay_gamma"])

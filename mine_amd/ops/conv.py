@@ -116,7 +116,7 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
     # is not paid back.
     usable = (x.is_cuda and x.dtype == torch.bfloat16
               and x.shape[1] % 8 == 0 and x.shape[1] <= 64
-              and x.shape[-1] >= 48
+              and x.shape[-1] >= 48 and x.shape[-2] >= 8
               and x.is_contiguous(memory_format=torch.channels_last))
     if usable:
         return _Conv3x3ReflFn.apply(x, w, bias)

@@ -48,11 +48,12 @@ def init_distributed(backend: Optional[str] = None) -> tuple:
 
 
 class _Bucket:
-    __slots__ = ("flat", "params", "ready", "handle", "offsets")
+    __slots__ = ("flat", "comm", "params", "ready", "handle", "offsets")
 
     def __init__(self, flat: torch.Tensor, params: List[torch.nn.Parameter],
-                 offsets: List[int]):
+                 offsets: List[int], comm: Optional[torch.Tensor] = None):
         self.flat = flat
+        self.comm = comm  # reduced-precision wire buffer (None = reduce flat)
         self.params = params
         self.offsets = offsets
         self.ready = 0
@@ -72,10 +73,16 @@ class GradAllReduceEngine:
     def __init__(self, modules: Iterable[torch.nn.Module],
                  bucket_mb: float = 25.0,
                  process_group: Optional[dist.ProcessGroup] = None,
-                 broadcast_params: bool = True):
+                 broadcast_params: bool = True,
+                 allreduce_dtype: Optional[torch.dtype] = None):
+        """allreduce_dtype: wire dtype for the gradient collectives
+        (e.g. torch.bfloat16 halves the xGMI bytes; grads still
+        ACCUMULATE in the fp32 buckets — only the reduction itself is
+        compressed). None = reduce the fp32 buckets directly."""
         self.pg = process_group
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.enabled = self.world_size > 1
+        self.allreduce_dtype = allreduce_dtype
 
         params: List[torch.nn.Parameter] = []
         for m in modules:
@@ -106,12 +113,17 @@ class GradAllReduceEngine:
                 return
             device = cur_params[0].device
             flat = torch.zeros(cur_numel, dtype=torch.float32, device=device)
+            comm = None
+            if self.enabled and self.allreduce_dtype is not None and \
+                    self.allreduce_dtype != torch.float32:
+                comm = torch.zeros(cur_numel, dtype=self.allreduce_dtype,
+                                   device=device)
             offsets = []
             off = 0
             for p in cur_params:
                 offsets.append(off)
                 off += p.numel()
-            b = _Bucket(flat, cur_params, offsets)
+            b = _Bucket(flat, cur_params, offsets, comm)
             self.buckets.append(b)
             for p, o in zip(cur_params, offsets):
                 self._param_bucket[id(p)] = (b, o)
@@ -141,12 +153,20 @@ class GradAllReduceEngine:
                     p.register_post_accumulate_grad_hook(self._on_grad_ready))
 
     # ------------------------------------------------------------------
+    def _launch(self, b: "_Bucket") -> None:
+        if b.comm is not None:
+            b.comm.copy_(b.flat)
+            b.handle = dist.all_reduce(b.comm, op=dist.ReduceOp.SUM,
+                                       group=self.pg, async_op=True)
+        else:
+            b.handle = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                       group=self.pg, async_op=True)
+
     def _on_grad_ready(self, p: torch.nn.Parameter) -> None:
         b, _ = self._param_bucket[id(p)]
         b.ready += 1
         if b.ready == len(b.params):
-            b.handle = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
-                                       group=self.pg, async_op=True)
+            self._launch(b)
 
     def zero_grad(self) -> None:
         for b in self.buckets:
@@ -162,11 +182,12 @@ class GradAllReduceEngine:
             return
         for b in self.buckets:
             if b.handle is None:
-                b.handle = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
-                                           group=self.pg, async_op=True)
+                self._launch(b)
         inv = 1.0 / self.world_size
         for b in self.buckets:
             b.handle.wait()
+            if b.comm is not None:
+                b.flat.copy_(b.comm)
             b.flat.mul_(inv)
             b.handle = None
             b.ready = 0

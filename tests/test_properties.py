@@ -121,3 +121,30 @@ def test_inverse_3x3_random_wellconditioned(seed):
     torch.testing.assert_close(torch.matmul(A, Ai),
                                torch.eye(3).expand(4, 3, 3),
                                rtol=1e-3, atol=1e-3)
+
+
+def test_reflect_conv_backward_data_identity():
+    """Math groundwork for the round-2 custom bwd-data kernel
+    (docs/NEXT.md #1): for y = conv_valid(reflect_pad1(x), W),
+    dL/dx == reflect_fold( conv_zero_pad2(gy, rot180(W).swapdims(0,1)) )
+    where reflect_fold is the pad-gradient gather. Verified against
+    autograd on CPU."""
+    import torch.nn.functional as F
+
+    g = torch.Generator().manual_seed(3)
+    B, C, K, H, W = 2, 3, 5, 6, 9
+    x = torch.randn(B, C, H, W, generator=g, requires_grad=True)
+    w = torch.randn(K, C, 3, 3, generator=g)
+    gy = torch.randn(B, K, H, W, generator=g)
+
+    y = F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"), w)
+    (y * gy).sum().backward()
+
+    # candidate formulation
+    w_t = w.permute(1, 0, 2, 3).flip(2, 3)          # (C, K, 3, 3)
+    gxp = F.conv2d(F.pad(gy, (2, 2, 2, 2)), w_t)    # (B, C, H+2, W+2)
+    # reflect_fold: transpose of reflect-pad — same gather the HIP
+    # pad-backward kernel implements
+    xp_probe = torch.zeros(B, C, H, W, requires_grad=True)
+    F.pad(xp_probe, (1, 1, 1, 1), mode="reflect").backward(gxp)
+    torch.testing.assert_close(xp_probe.grad, x.grad, rtol=1e-4, atol=1e-5)

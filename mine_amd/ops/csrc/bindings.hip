@@ -35,6 +35,8 @@ void mine_reflect_pad_bwd_bf16(const void*, void*, int, int, int, int, int,
                                hipStream_t);
 void mine_conv3x3_fwd(const void*, const void*, const float*, void*, int,
                       int, int, int, int, int, hipStream_t);
+void mine_conv3x3_wrw(const void*, const void*, float*, int, int, int, int,
+                      int, hipStream_t);
 void mine_mpi_head_fwd_f32(const void*, float*, int64_t, int, hipStream_t);
 void mine_mpi_head_fwd_bf16(const void*, float*, int64_t, int, hipStream_t);
 void mine_mpi_head_bwd_f32(const void*, const float*, void*, int64_t, int,
@@ -225,6 +227,22 @@ void conv3x3_fwd(at::Tensor x_flat, at::Tensor wp, at::Tensor bias,
                    bias.numel() ? bias.data_ptr<float>() : nullptr,
                    out.data_ptr(), (int)N, (int)H, (int)W, (int)C, (int)K,
                    (int)pad_mode, stream());
+}
+
+// EXPERIMENTAL split-K MFMA weight gradient (see wrw_kernels.hip);
+// returns dW fp32 (K, C, 3, 3). Flat NHWC bf16 inputs.
+at::Tensor conv3x3_wrw(at::Tensor x_flat, at::Tensor gy_flat, int64_t N,
+                       int64_t H, int64_t W, int64_t C, int64_t K) {
+  TORCH_CHECK(x_flat.is_cuda() && x_flat.is_contiguous() &&
+              gy_flat.is_contiguous());
+  TORCH_CHECK(x_flat.scalar_type() == at::kBFloat16 &&
+              gy_flat.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(C % 8 == 0 && C <= 32 && K <= 32 && W <= 1022);
+  auto dw = at::zeros({K, C, 3, 3}, x_flat.options().dtype(at::kFloat));
+  mine_conv3x3_wrw(x_flat.data_ptr(), gy_flat.data_ptr(),
+                   dw.data_ptr<float>(), (int)N, (int)H, (int)W, (int)C,
+                   (int)K, stream());
+  return dw;
 }
 
 // fused MPI head over a flat (N,4) view; out is fp32 (N,4)
@@ -429,4 +447,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("mpi_head_bwd", &mpi_head_bwd);
   mod.def("conv3x3_fwd", &conv3x3_fwd,
           "fused reflect/zero-pad + 3x3 conv on MFMA 16x16x32 tiles");
+  mod.def("conv3x3_wrw", &conv3x3_wrw,
+          "EXPERIMENTAL split-K MFMA weight gradient (unwired)");
 }

@@ -1,4 +1,5 @@
 """HIP kernels vs the pure-torch oracle (runs on an MI355X box)."""
+import os
 import pytest
 import torch
 
@@ -464,3 +465,32 @@ def test_mfma_conv3x3_speed_vs_miopen():
     print(f"\n[conv3x3 {B}x{C}x{H}x{W}->{K}] pad+MIOpen {t_ref:.3f} ms, "
           f"fused MFMA {t_mfma:.3f} ms, speedup {t_ref / t_mfma:.2f}x")
     assert t_mfma < t_ref * 1.5  # must at least be in the same class
+
+
+@pytest.mark.skipif(os.environ.get("MINE_EXPERIMENTAL") != "1",
+                    reason="experimental kernel (round-2 groundwork); "
+                           "set MINE_EXPERIMENTAL=1 to run")
+@pytest.mark.parametrize("shape", [(3, 16, 12, 50, 16), (2, 32, 9, 64, 16),
+                                   (2, 16, 20, 70, 4)])
+def test_experimental_wrw_matches_torch(shape):
+    import torch.nn.functional as F
+    from mine_amd.ops.backend import get_extension
+
+    B, C, H, W, K = shape
+    g = torch.Generator().manual_seed(17)
+    x = torch.randn(B, C, H, W, generator=g).to("cuda:0", torch.bfloat16
+        ).contiguous(memory_format=torch.channels_last)
+    gy = torch.randn(B, K, H, W, generator=g).to("cuda:0", torch.bfloat16
+        ).contiguous(memory_format=torch.channels_last)
+
+    ext = get_extension(required=True)
+    dw = ext.conv3x3_wrw(x.permute(0, 2, 3, 1).reshape(-1),
+                         gy.permute(0, 2, 3, 1).reshape(-1), B, H, W, C, K)
+
+    xq = x.float().cpu().requires_grad_(False)
+    xp = F.pad(xq, (1, 1, 1, 1), mode="reflect")
+    w_probe = torch.zeros(K, C, 3, 3, requires_grad=True)
+    y = F.conv2d(xp, w_probe)
+    (y * gy.float().cpu()).sum().backward()
+
+    torch.testing.assert_close(dw.cpu(), w_probe.grad, rtol=2e-2, atol=2e-1)

@@ -76,7 +76,7 @@ class _Conv3x3ReflFn(torch.autograd.Function):
         ext.conv3x3_fwd(x_flat, wp,
                         bias.float() if bias is not None else
                         torch.empty(0, device=x.device, dtype=torch.float32),
-                        out, B, H, W, C, K, 0)
+                        out, B, H, W, C, K, 0, H, W, 0)
         # save the PADDED input for backward (one cheap HIP pad; MIOpen's
         # tuned convolution_backward then runs exactly as in the
         # unfused path — no recompute in the backward hot loop)
@@ -123,3 +123,29 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
     from mine_amd.ops.pad import reflection_pad2d
     return F.conv2d(reflection_pad2d(x, 1), w.to(x.dtype),
                     bias.to(x.dtype) if bias is not None else None)
+
+
+def conv3x3_bwd_data_experimental(gy: torch.Tensor,
+                                  w: torch.Tensor) -> torch.Tensor:
+    """EXPERIMENTAL (round-2; unwired): data gradient of the fused
+    reflect-pad conv via the SAME MFMA kernel in zero-embed mode —
+    gx = reflect_fold( conv_zero_pad2(gy, rot180(W).swap(0,1)) ), the
+    identity verified on CPU in tests/test_properties.py.
+
+    gy: (B, K, H, W) bf16 channels_last; w: (K, C, 3, 3). Returns
+    gx (B, C, H, W) bf16 channels_last.
+    """
+    ext = get_extension(required=True)
+    B, K, H, W = gy.shape
+    C = w.shape[1]
+    w_t = w.permute(1, 0, 2, 3).flip(2, 3).contiguous()  # (C, K, 3, 3)
+    wp = pack_weights(w_t.to(torch.bfloat16))
+    gy_flat = gy.permute(0, 2, 3, 1).reshape(-1)
+    # logical image = gy zero-embedded by 1 ring -> output (H+2, W+2, C)
+    gxp = torch.empty(B * (H + 2) * (W + 2) * C, device=gy.device,
+                      dtype=torch.bfloat16)
+    ext.conv3x3_fwd(gy_flat, wp,
+                    torch.empty(0, device=gy.device, dtype=torch.float32),
+                    gxp, B, H + 2, W + 2, K, C, 1, H, W, 1)
+    gx = ext.reflect_pad_bwd(gxp, B, H, W, C, 1)
+    return gx.view(B, H, W, C).permute(0, 3, 1, 2)

@@ -34,7 +34,7 @@ void mine_reflect_pad_bwd_f32(const float*, float*, int, int, int, int, int,
 void mine_reflect_pad_bwd_bf16(const void*, void*, int, int, int, int, int,
                                hipStream_t);
 void mine_conv3x3_fwd(const void*, const void*, const float*, void*, int,
-                      int, int, int, int, int, hipStream_t);
+                      int, int, int, int, int, int, int, int, hipStream_t);
 void mine_conv3x3_wrw(const void*, const void*, float*, int, int, int, int,
                       int, hipStream_t);
 void mine_mpi_head_fwd_f32(const void*, float*, int64_t, int, hipStream_t);
@@ -213,20 +213,25 @@ at::Tensor reflect_pad_bwd(at::Tensor gout, int64_t N, int64_t H, int64_t W,
     else TORCH_CHECK(false, "bn: dtype must be f32 or bf16");     \
   } while (0)
 
-// fused reflect-pad + 3x3 conv (MFMA); flat NHWC views, bf16
+// fused reflect-pad + 3x3 conv (MFMA); flat NHWC views, bf16.
+// (src_h, src_w, off) back a zero-embedded logical image for the
+// transposed/data-grad use (pad_mode 1, off 1, logical = src + 2);
+// plain forward: src == logical, off == 0.
 void conv3x3_fwd(at::Tensor x_flat, at::Tensor wp, at::Tensor bias,
                  at::Tensor out, int64_t N, int64_t H, int64_t W, int64_t C,
-                 int64_t K, int64_t pad_mode) {
+                 int64_t K, int64_t pad_mode, int64_t src_h, int64_t src_w,
+                 int64_t off) {
   TORCH_CHECK(x_flat.is_cuda() && x_flat.is_contiguous());
   TORCH_CHECK(x_flat.scalar_type() == at::kBFloat16 &&
               wp.scalar_type() == at::kBFloat16 &&
               out.scalar_type() == at::kBFloat16);
-  TORCH_CHECK(C % 8 == 0 && x_flat.numel() == N * H * W * C);
+  TORCH_CHECK(C % 8 == 0 && x_flat.numel() == N * src_h * src_w * C);
   TORCH_CHECK(out.numel() == N * H * W * K);
   mine_conv3x3_fwd(x_flat.data_ptr(), wp.data_ptr(),
                    bias.numel() ? bias.data_ptr<float>() : nullptr,
                    out.data_ptr(), (int)N, (int)H, (int)W, (int)C, (int)K,
-                   (int)pad_mode, stream());
+                   (int)pad_mode, (int)src_h, (int)src_w, (int)off,
+                   stream());
 }
 
 // EXPERIMENTAL split-K MFMA weight gradient (see wrw_kernels.hip);

@@ -41,23 +41,28 @@ constexpr int STAGE_H = TILE_H + 2; // 6 staged input rows (halo shared)
 enum PadMode { PAD_REFLECT = 0, PAD_ZERO = 1 };
 
 template <int PAD>
-__device__ __forceinline__ int map_coord(int v, int n) {
-  // reflect: -1 -> 1, n -> n-2 (pad 1); zero: OOB -> -1 sentinel
+__device__ __forceinline__ int map_coord(int v, int n, int src_n, int off) {
+  // reflect: -1 -> 1, n -> n-2 (pad 1); zero: OOB -> -1 sentinel.
+  // (src_n, off) give the backing array's extent when the LOGICAL image
+  // is a zero-embedded ring around it (the transposed-conv use: logical
+  // H = src_H + 2, off = 1); for the plain forward src_n == n, off == 0.
   if (PAD == PAD_REFLECT) {
     if (v < 0) v = -v;
     if (v >= n) v = 2 * (n - 1) - v;
     return v;
   }
-  return (v < 0 || v >= n) ? -1 : v;
+  v -= off;
+  return (v < 0 || v >= src_n) ? -1 : v;
 }
 
 template <int PAD>
 __global__ void __launch_bounds__(kBlock)
-conv3x3_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,H,W,C)
+conv3x3_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,sH,sW,C)
                    const __hip_bfloat16* __restrict__ wp, // packed frags
                    const float* __restrict__ bias,        // (K) or null
                    __hip_bfloat16* __restrict__ out,      // (N,H,W,K)
-                   int H, int W, int C, int K) {
+                   int H, int W, int C, int K,
+                   int sH, int sW, int off) {
   extern __shared__ __hip_bfloat16 s_in[];  // [STAGE_H][STAGE_W][C]
   const int x0 = blockIdx.x * TILE_W;
   const int y0 = blockIdx.y * TILE_H;
@@ -67,21 +72,21 @@ conv3x3_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,H,W,C)
   // ---- stage 6 reflected rows x 66 pixels x C (8-channel vectors) ----
   const int Cv = C / 8;
   const int total_v = STAGE_H * STAGE_W * Cv;
-  const int64_t x_n = (int64_t)n * H * W * C;
+  const int64_t x_n = (int64_t)n * sH * sW * C;
   for (int i = tid; i < total_v; i += kBlock) {
     const int cv = i % Cv;
     const int rem = i / Cv;
     const int sx = rem % STAGE_W;      // 0..65 -> input x = x0 + sx - 1
     const int row = rem / STAGE_W;     // 0..5  -> input y = y0 + row - 1
-    const int yy = map_coord<PAD>(y0 + row - 1, H);
-    const int xx = map_coord<PAD>(x0 + sx - 1, W);
+    const int yy = map_coord<PAD>(y0 + row - 1, H, sH, off);
+    const int xx = map_coord<PAD>(x0 + sx - 1, W, sW, off);
     bf16x8 v;
     if (yy < 0 || xx < 0) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) v[e] = (__bf16)0.0f;
     } else {
       v = *reinterpret_cast<const bf16x8*>(
-          x + x_n + ((int64_t)yy * W + xx) * C + cv * 8);
+          x + x_n + ((int64_t)yy * sW + xx) * C + cv * 8);
     }
     *reinterpret_cast<bf16x8*>(s_in + (i * 8)) = v;
   }
@@ -160,6 +165,7 @@ conv3x3_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,H,W,C)
 extern "C" void mine_conv3x3_fwd(const void* x, const void* wp,
                                  const float* bias, void* out, int N, int H,
                                  int W, int C, int K, int pad_mode,
+                                 int src_h, int src_w, int off,
                                  hipStream_t stream) {
   const dim3 grid((W + TILE_W - 1) / TILE_W, (H + TILE_H - 1) / TILE_H, N);
   const size_t lds = STAGE_H * STAGE_W * C * sizeof(__hip_bfloat16);
@@ -168,11 +174,13 @@ extern "C" void mine_conv3x3_fwd(const void* x, const void* wp,
                        lds, stream,
                        reinterpret_cast<const __hip_bfloat16*>(x),
                        reinterpret_cast<const __hip_bfloat16*>(wp), bias,
-                       reinterpret_cast<__hip_bfloat16*>(out), H, W, C, K);
+                       reinterpret_cast<__hip_bfloat16*>(out), H, W, C, K,
+                       src_h, src_w, off);
   else
     hipLaunchKernelGGL(conv3x3_fwd_kernel<PAD_ZERO>, grid, dim3(kBlock),
                        lds, stream,
                        reinterpret_cast<const __hip_bfloat16*>(x),
                        reinterpret_cast<const __hip_bfloat16*>(wp), bias,
-                       reinterpret_cast<__hip_bfloat16*>(out), H, W, C, K);
+                       reinterpret_cast<__hip_bfloat16*>(out), H, W, C, K,
+                       src_h, src_w, off);
 }

@@ -494,3 +494,27 @@ def test_experimental_wrw_matches_torch(shape):
     (y * gy.float().cpu()).sum().backward()
 
     torch.testing.assert_close(dw.cpu(), w_probe.grad, rtol=2e-2, atol=2e-1)
+
+
+@pytest.mark.skipif(os.environ.get("MINE_EXPERIMENTAL") != "1",
+                    reason="experimental kernel (round-2 groundwork)")
+def test_experimental_bwd_data_matches_autograd():
+    import torch.nn.functional as F
+    from mine_amd.ops.conv import conv3x3_bwd_data_experimental
+
+    B, C, H, W, K = 2, 16, 10, 54, 16
+    g = torch.Generator().manual_seed(23)
+    x = torch.randn(B, C, H, W, generator=g)
+    w = torch.randn(K, C, 3, 3, generator=g) * 0.2
+    gy = torch.randn(B, K, H, W, generator=g)
+
+    gx = conv3x3_bwd_data_experimental(
+        gy.to("cuda:0", torch.bfloat16).contiguous(
+            memory_format=torch.channels_last),
+        w.cuda())
+
+    xr = x.clone().requires_grad_(True)
+    y = F.conv2d(F.pad(xr.to(torch.bfloat16).float(), (1, 1, 1, 1),
+                       mode="reflect"), w, None)
+    (y * gy.to(torch.bfloat16).float()).sum().backward()
+    torch.testing.assert_close(gx.float().cpu(), xr.grad, rtol=5e-2, atol=5e-2)

@@ -123,12 +123,20 @@ class SynthesisTask:
              {"params": self.decoder.parameters(), "lr": config["lr.decoder_lr"]}],
             weight_decay=config["lr.weight_decay"])
 
-        # rank-0 restore BEFORE the parameter broadcast (ref CS5)
+        # rank-0 restore BEFORE the parameter broadcast (ref CS5).
+        # "auto" resumes from the workspace's checkpoint_latest.pth when
+        # one exists — an elastic-restart convenience the reference
+        # lacked (it always restarted at epoch 1).
+        ckpt_path = config.get("training.pretrained_checkpoint_path")
+        if ckpt_path == "auto":
+            cand = os.path.join(self.state.local_workspace or "",
+                                "checkpoint_latest.pth")
+            ckpt_path = cand if os.path.exists(cand) else None
         self._restored_meta = {}
         if self.state.is_rank0:
             self._restored_meta = restore_model(
-                config.get("training.pretrained_checkpoint_path"),
-                self.backbone, self.decoder, self.optimizer, logger=logger) or {}
+                ckpt_path, self.backbone, self.decoder, self.optimizer,
+                logger=logger) or {}
 
         self.grad_engine = None
         if not is_val:

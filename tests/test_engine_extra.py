@@ -173,3 +173,28 @@ def test_training_reduces_loss():
         losses.append(float(task.train_step(items)["loss"]))
     last = sum(losses[-5:]) / 5
     assert last < 0.7 * first, (first, losses[-5:])
+
+
+def test_auto_resume_from_workspace(tmp_path):
+    from mine_amd.config import RuntimeState
+
+    cfg = _cfg()
+    task = SynthesisTask(cfg, device="cpu")
+    task.current_epoch = 2
+    task.global_step = 77
+    save_checkpoint(str(tmp_path / "checkpoint_latest.pth"),
+                    task.backbone, task.decoder, task.optimizer,
+                    meta={"epoch": 2, "global_step": 77})
+
+    st = RuntimeState(local_workspace=str(tmp_path))
+    cfg2 = cfg.replace(**{"training.pretrained_checkpoint_path": "auto"})
+    task2 = SynthesisTask(cfg2, state=st, device="cpu")
+    assert task2._restored_meta.get("global_step") == 77
+    torch.testing.assert_close(
+        task2.decoder.dispconvs["0"].conv.weight,
+        task.decoder.dispconvs["0"].conv.weight)
+
+    # "auto" with an empty workspace: cold start, no error
+    st3 = RuntimeState(local_workspace=str(tmp_path / "empty"))
+    task3 = SynthesisTask(cfg2, state=st3, device="cpu")
+    assert task3._restored_meta == {}

@@ -1,0 +1,144 @@
+"""CPU simulations of the experimental MFMA kernels' index math.
+
+The HIP sources (ops/csrc/wrw_kernels.hip, the zero-embed mode of
+conv_kernels.hip) are round-2 groundwork that cannot be executed here;
+these tests re-enact their exact staging + fragment addressing + MFMA
+contraction in torch and check the result against autograd, so any edit
+that breaks the index math fails HERE before it ever reaches a GPU.
+"""
+import os
+import sys
+
+import torch
+import torch.nn.functional as F
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
+
+def _reflect1(v, n):
+    if v < 0:
+        v = -v
+    if v >= n:
+        v = 2 * (n - 1) - v
+    return v
+
+
+def test_wrw_kernel_index_math():
+    """Simulates conv3x3_wrw_kernel: slab split, gy^T / x^T staging with
+    reflect halo, 32-pixel contraction chunks, (c*9+tap) column order."""
+    torch.manual_seed(4)
+    B, C, H, W, K = 2, 8, 5, 40, 16
+    x = torch.randn(B, C, H, W)
+    gy = torch.randn(B, K, H, W)
+
+    Cv = C // 8
+    nch = (9 * Cv + 1) // 2
+    px_chunks = (W + 31) // 32
+    n_slabs = min(512, B * H)
+    xn = x.permute(0, 2, 3, 1).double()
+    gyn = gy.permute(0, 2, 3, 1).double()
+    dw = torch.zeros(K, 9 * C, dtype=torch.float64)
+
+    for kc in range((K + 15) // 16):
+        k0 = kc * 16
+        for slab in range(n_slabs):
+            r0, r1 = B * H * slab // n_slabs, B * H * (slab + 1) // n_slabs
+            part = torch.zeros(16, nch * 16, dtype=torch.float64)
+            for r in range(r0, r1):
+                n, y = divmod(r, H)
+                s_gy = torch.zeros(16, W, dtype=torch.float64)
+                for k in range(16):
+                    if k0 + k < K:
+                        s_gy[k] = gyn[n, y, :, k0 + k]
+                s_x = torch.zeros(C, 3, W + 2, dtype=torch.float64)
+                for c in range(C):
+                    for row in range(3):
+                        yy = _reflect1(y + row - 1, H)
+                        for xx in range(W + 2):
+                            s_x[c, row, xx] = xn[n, yy, _reflect1(xx - 1, W), c]
+                for pc in range(px_chunks):
+                    p0 = pc * 32
+                    A = torch.zeros(16, 32, dtype=torch.float64)
+                    Bm = torch.zeros(32, nch * 16, dtype=torch.float64)
+                    for p in range(32):
+                        if p0 + p < W:
+                            A[:, p] = s_gy[:, p0 + p]
+                    for col in range(nch * 16):
+                        if col < 9 * C:
+                            c, tap = divmod(col, 9)
+                            dy, dx = divmod(tap, 3)
+                            for p in range(32):
+                                if p0 + p < W:
+                                    Bm[p, col] = s_x[c, dy, p0 + p + dx]
+                    part += A @ Bm
+            dw[k0:k0 + min(16, K - k0)] += part[:min(16, K - k0), :9 * C]
+    dw = dw.view(K, C, 3, 3).float()
+
+    w_probe = torch.zeros(K, C, 3, 3, requires_grad=True)
+    y = F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"), w_probe)
+    (y * gy).sum().backward()
+    torch.testing.assert_close(dw, w_probe.grad, rtol=1e-4, atol=1e-4)
+
+
+def test_fwd_kernel_pack_and_zero_embed_math():
+    """Simulates conv3x3_fwd_kernel addressing for BOTH pad modes:
+    (a) the (cb*9+tap)*8+ci k-ordering of pack_weights against reflect
+    conv; (b) the zero-embed logical extent (src_h, src_w, off=1) used by
+    the experimental bwd-data path."""
+    from mine_amd.ops.conv import _pack_lut
+
+    torch.manual_seed(6)
+    C, K, H, W = 8, 16, 4, 9
+    x = torch.randn(1, C, H, W)
+    w = torch.randn(K, C, 3, 3)
+
+    # (a) reconstruct the GEMM from the pack LUT + per-pixel A rows
+    lut = _pack_lut(K, C, False, torch.device("cpu"))
+    Cv = C // 8
+    nseg = 9 * Cv
+    nchunks = (nseg + 3) // 4
+    nK = (K + 15) // 16
+    flat = torch.cat((w.reshape(-1), torch.zeros(1)))
+    wp = flat[lut].view(nK, nchunks, 64, 8)
+
+    xp = F.pad(x, (1, 1, 1, 1), mode="reflect").permute(0, 2, 3, 1)[0]  # H+2,W+2,C
+    out = torch.zeros(H, W, K)
+    for yy in range(H):
+        for xx in range(W):
+            for nc in range(nK):
+                for kc in range(nchunks):
+                    for lane in range(64):
+                        j = lane & 15
+                        seg = kc * 4 + (lane >> 4)
+                        kout = nc * 16 + j
+                        if seg >= nseg or kout >= K:
+                            continue
+                        cb, tap = divmod(seg, 9)
+                        dy, dx = divmod(tap, 3)
+                        for e in range(8):
+                            a = xp[yy + dy, xx + dx, cb * 8 + e]
+                            out[yy, xx, kout] += float(a) * float(wp[nc, kc, lane, e])
+    ref = F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"), w)[0].permute(1, 2, 0)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-4)
+
+    # (b) zero-embed: logical (H+2, W+2) image backed by (H, W) with off=1
+    def map_zero(v, n_logical, src_n, off):
+        v2 = v - off
+        return v2 if 0 <= v2 < src_n else None
+
+    Hl, Wl = H + 2, W + 2
+    gsrc = x.permute(0, 2, 3, 1)[0]
+    outz = torch.zeros(Hl, Wl, K)
+    for yy in range(Hl):
+        for xx in range(Wl):
+            for dy in range(3):
+                for dx in range(3):
+                    ys = map_zero(yy + dy - 1, Hl, H, 1)
+                    xs = map_zero(xx + dx - 1, Wl, W, 1)
+                    if ys is None or xs is None:
+                        continue
+                    for k in range(K):
+                        outz[yy, xx, k] += float(
+                            (gsrc[ys, xs] * w[k, :, dy, dx]).sum())
+    refz = F.conv2d(F.pad(x, (2, 2, 2, 2)), w)[0].permute(1, 2, 0)
+    torch.testing.assert_close(outz, refz, rtol=1e-4, atol=1e-4)

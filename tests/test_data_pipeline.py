@@ -303,3 +303,33 @@ def test_eval_pairs_protocol(tmp_path):
     assert out["5_frames"]["n_pairs"] == 1
     assert out["10_frames"]["n_pairs"] == 1
     assert out["5_frames"]["psnr"] > 0
+
+
+def test_llff_validation_split_folder(tmp_path):
+    """Validation reads images_<ratio>_val when present (ref
+    nerf_dataset.py:47-53) and falls back to the train folder otherwise."""
+    import shutil
+    from mine_amd.config import default_config
+    from mine_amd.data.llff import NeRFDataset
+
+    _make_scene(str(tmp_path))
+    scene = os.path.join(str(tmp_path), "scene0")
+    # dedicated val folder with a SUBSET of views
+    val_dir = os.path.join(scene, "images_7.875_val")
+    os.makedirs(val_dir)
+    for name in sorted(os.listdir(os.path.join(scene, "images_7.875")))[:3]:
+        shutil.copy(os.path.join(scene, "images_7.875", name),
+                    os.path.join(val_dir, name))
+
+    cfg = default_config(**{
+        "data.name": "llff", "data.img_h": 32, "data.img_w": 40,
+        "data.visible_point_count": 8,
+        "data.training_set_path": str(tmp_path)})
+    val = NeRFDataset(cfg, root=str(tmp_path), is_validation=True,
+                      img_size=(40, 32), visible_points_count=8)
+    assert len(val) == 3  # only the val-folder views
+    # deterministic target pick in validation
+    _, t1 = val[0]
+    _, t2 = val[0]
+    torch.testing.assert_close(t1[0]["img"], t2[0]["img"])
+    torch.testing.assert_close(t1[0]["G_src_tgt"], t2[0]["G_src_tgt"])

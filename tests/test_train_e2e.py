@@ -70,3 +70,30 @@ def test_bench_distributed_torchrun_cpu():
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"] == "dp2"
     assert out["value"] > 0
+
+
+def test_train_cli_two_ranks_gloo(tmp_path):
+    """train.py under torchrun with 2 CPU ranks: DistributedSampler,
+    rank-0 logging/checkpointing, barrier, grad engine."""
+    extra = {
+        "data.name": "synthetic", "data.img_h": 64, "data.img_w": 64,
+        "mpi.num_bins_coarse": 4, "data.per_gpu_batch_size": 2,
+        "data.visible_point_count": 8, "training.amp_dtype": "fp32",
+        "data.synthetic_length": 8, "training.epochs": 1,
+        "training.eval_interval": 1000000, "data.num_workers": 0,
+        "training.checkpoint_interval": 2,
+    }
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29613", os.path.join(ROOT, "train.py"),
+         "--config_path", os.path.join(ROOT, "configs", "params_default.yaml"),
+         "--workspace", str(tmp_path), "--version", "t2",
+         "--extra_config", json.dumps(extra)],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=900)
+    assert r.returncode == 0, (r.stdout[-1200:], r.stderr[-1500:])
+    ws = tmp_path / "t2"
+    assert (ws / "checkpoint_latest.pth").exists()
+    assert (ws / "training.log").exists()

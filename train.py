@@ -59,6 +59,17 @@ def main():
     state = RuntimeState(global_rank=rank, local_rank=local_rank,
                          world_size=world_size)
 
+    # Device selection: when training.gpus masked CUDA_VISIBLE_DEVICES
+    # above, every rank sees its GPU as cuda:0; otherwise (e.g. default
+    # `training.gpus: 0` under torchrun with all GPUs visible) each rank
+    # takes device local_rank — without this, ranks 1+ would all land on
+    # cuda:0.
+    device = None
+    if torch.cuda.is_available():
+        dev_idx = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev_idx)
+        device = f"cuda:{dev_idx}"
+
     workspace = os.path.join(args.workspace, args.version)
     state.local_workspace = workspace
     logger = None
@@ -94,7 +105,7 @@ def main():
                             shuffle=False, drop_last=False, num_workers=0,
                             collate_fn=val_dataset.collate_fn)
 
-    task = SynthesisTask(config, state=state, logger=logger)
+    task = SynthesisTask(config, state=state, logger=logger, device=device)
     task.train(train_loader, val_loader)
 
 

@@ -142,3 +142,78 @@ def test_fwd_kernel_pack_and_zero_embed_math():
                             (gsrc[ys, xs] * w[k, :, dy, dx]).sum())
     refz = F.conv2d(F.pad(x, (2, 2, 2, 2)), w)[0].permute(1, 2, 0)
     torch.testing.assert_close(outz, refz, rtol=1e-4, atol=1e-4)
+
+
+def test_warp_backward_gather_decomposition():
+    """Round-2 design validation (docs/NEXT.md #2): the tgt-composite
+    backward's bilinear scatter equals an INTERIOR per-src-pixel gather
+    (inverse-homography window search) plus a rare-path scatter for the
+    border-clamped tgt pixels. Proven here in torch before any kernel."""
+    from mine_amd.ops import torch_ref as tr
+
+    def make_scene(seed, B=1, S=3, H=14, W=18):
+        g = torch.Generator().manual_seed(seed)
+        disparity, _ = torch.sort(torch.rand(B, S, generator=g) * 0.9 + 0.05,
+                                  dim=1, descending=True)
+        f = 0.8 * W
+        K = torch.tensor([[f, 0, W / 2], [0, f, H / 2], [0, 0, 1.0]]).unsqueeze(0)
+        aa = 0.1 * torch.randn(3, generator=g)
+        th = aa.norm()
+        k = aa / (th + 1e-9)
+        Kx = torch.tensor([[0, -k[2], k[1]], [k[2], 0, -k[0]],
+                           [-k[1], k[0], 0.0]])
+        R = torch.eye(3) + th.sin() * Kx + (1 - th.cos()) * (Kx @ Kx)
+        G = torch.eye(4).unsqueeze(0).clone()
+        G[0, :3, :3] = R
+        G[0, :3, 3] = 0.2 * torch.randn(3, generator=g)
+        payload = torch.randn(B, S, H, W, 4, generator=g)
+        return disparity, K, torch.inverse(K), G, payload
+
+    def taps(u, v, H, W):
+        u = min(max(u, 0.0), W - 1)
+        v = min(max(v, 0.0), H - 1)
+        x0, y0 = int(u), int(v)
+        x1, y1 = min(x0 + 1, W - 1), min(y0 + 1, H - 1)
+        wx, wy = u - x0, v - y0
+        return ((x0, y0, (1 - wx) * (1 - wy)), (x1, y0, wx * (1 - wy)),
+                (x0, y1, (1 - wx) * wy), (x1, y1, wx * wy))
+
+    for seed in range(3):
+        disparity, K, K_inv, G, payload = make_scene(seed)
+        B, S, H, W = payload.shape[:4]
+        depths = torch.reciprocal(disparity)
+        Hinv = tr.homography_tgt_to_src(G, depths, K_inv, K)
+        grid = tr.make_meshgrid(H, W)
+        uvh = torch.einsum("bsij,jhw->bsihw", Hinv, grid)
+        uv = uvh[:, :, :2] / uvh[:, :, 2:]
+
+        scatter = torch.zeros(B, S, H, W, 4)
+        gather = torch.zeros(B, S, H, W, 4)
+        for b in range(B):
+            for s in range(S):
+                U, V = uv[b, s, 0], uv[b, s, 1]
+                interior = (U >= 0) & (U <= W - 1) & (V >= 0) & (V <= H - 1)
+                for y in range(H):
+                    for x in range(W):
+                        for qx, qy, wt in taps(float(U[y, x]), float(V[y, x]),
+                                               H, W):
+                            scatter[b, s, qy, qx] += wt * payload[b, s, y, x]
+                        if not interior[y, x]:  # rare path in the gather form
+                            for qx, qy, wt in taps(float(U[y, x]),
+                                                   float(V[y, x]), H, W):
+                                gather[b, s, qy, qx] += wt * payload[b, s, y, x]
+                Hf = torch.inverse(Hinv[b, s])
+                for qy in range(H):
+                    for qx in range(W):
+                        pc = Hf @ torch.tensor([qx, qy, 1.0])
+                        px, py = float(pc[0] / pc[2]), float(pc[1] / pc[2])
+                        for ty in range(max(0, int(py) - 4), min(H, int(py) + 6)):
+                            for tx in range(max(0, int(px) - 4), min(W, int(px) + 6)):
+                                if not interior[ty, tx]:
+                                    continue
+                                du = float(U[ty, tx]) - qx
+                                dv = float(V[ty, tx]) - qy
+                                if -1 < du < 1 and -1 < dv < 1:
+                                    gather[b, s, qy, qx] += (1 - abs(du)) * \
+                                        (1 - abs(dv)) * payload[b, s, ty, tx]
+        torch.testing.assert_close(gather, scatter, rtol=1e-4, atol=1e-5)

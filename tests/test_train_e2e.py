@@ -97,3 +97,38 @@ def test_train_cli_two_ranks_gloo(tmp_path):
     ws = tmp_path / "t2"
     assert (ws / "checkpoint_latest.pth").exists()
     assert (ws / "training.log").exists()
+
+
+def test_checkpoint_dir_contract_train_to_video(tmp_path):
+    """The reference's checkpoint-dir contract (params.yaml next to
+    checkpoint.pth, consumed by the inference CLI; ref
+    image_to_video.py:272-278): train.py writes it, image_to_video.py
+    renders from it."""
+    extra = {
+        "data.name": "synthetic", "data.img_h": 64, "data.img_w": 64,
+        "mpi.num_bins_coarse": 4, "data.per_gpu_batch_size": 2,
+        "data.visible_point_count": 8, "training.amp_dtype": "fp32",
+        "data.synthetic_length": 4, "training.epochs": 1,
+        "training.eval_interval": 1000000, "data.num_workers": 0,
+        "training.checkpoint_interval": 2,
+    }
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "train.py"),
+         "--config_path", os.path.join(ROOT, "configs", "params_default.yaml"),
+         "--workspace", str(tmp_path), "--version", "v1",
+         "--extra_config", json.dumps(extra)],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    ckpt = tmp_path / "v1" / "checkpoint_latest.pth"
+    assert ckpt.exists() and (tmp_path / "v1" / "params.yaml").exists()
+
+    r2 = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "visualizations", "image_to_video.py"),
+         "--checkpoint_path", str(ckpt),
+         "--output_dir", str(tmp_path / "vid"), "--num_frames", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=600)
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    assert len(list((tmp_path / "vid" / "frames").glob("*.png"))) == 2

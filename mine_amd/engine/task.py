@@ -166,6 +166,16 @@ class SynthesisTask:
             self.backbone.eval()
             self.decoder.eval()
 
+        if self.is_gpu and bool(config.get("training.stream_overlap", False)):
+            # EXPERIMENTAL (off by default; docs/NEXT.md #4): run the
+            # SplitConvBlocks' batch-B base convs on a side HIP stream,
+            # overlapped with the batch-B*S dec convs.
+            from mine_amd.models.decoder import SplitConvBlock
+            side = torch.cuda.Stream()
+            for m in self.decoder.modules():
+                if isinstance(m, SplitConvBlock):
+                    m.side_stream = side
+
         self.use_alpha = bool(config.get("mpi.use_alpha", False))
         self.bg_depth_inf = bool(config.get("mpi.is_bg_depth_inf", False))
         self.src_rgb_blending = bool(config.get("training.src_rgb_blending", True))

@@ -83,21 +83,45 @@ class SplitConvBlock(nn.Module):
     def forward(self, x_dec, base, pe, B: int, S: int) -> torch.Tensor:
         w = self.conv.weight
         d, b = self.dec_ch, self.base_ch
-        # base part at batch B (bias lives here; it is S-invariant)
-        y_base = F.conv2d(self.pad(base), w[:, d:d + b], self.conv.bias)
-        K, Hb, Wb = y_base.shape[1:]
-        # PE part: conv of a spatially-constant field == channel bias
-        w_pe = w[:, d + b:].sum((2, 3))  # (K, E)
-        bias_pe = torch.matmul(pe.to(w_pe.dtype), w_pe.t())  # (B*S, K)
-        # broadcast-add in NHWC so the (B*S,K,H,W) result is channels_last
-        # without a transpose (conv outputs are NHWC on GPU already)
-        yb = y_base.permute(0, 2, 3, 1).unsqueeze(1) \
-            + bias_pe.view(B, S, 1, 1, K).to(y_base.dtype)
-        yb = yb.view(B * S, Hb, Wb, K)
+
+        # Optional side-stream overlap (docs/NEXT.md #4; OFF by default,
+        # enabled by SynthesisTask when training.stream_overlap is set):
+        # the batch-B base conv + PE bias are independent of the batch-B*S
+        # dec conv until the add, so they can run concurrently.
+        side = getattr(self, "side_stream", None)
+        use_side = side is not None and base.is_cuda
+        main = torch.cuda.current_stream() if use_side else None
+
+        def base_part():
+            # base part at batch B (bias lives here; it is S-invariant)
+            y_base = F.conv2d(self.pad(base), w[:, d:d + b], self.conv.bias)
+            K = y_base.shape[1]
+            # PE part: conv of a spatially-constant field == channel bias
+            w_pe = w[:, d + b:].sum((2, 3))  # (K, E)
+            bias_pe = torch.matmul(pe.to(w_pe.dtype), w_pe.t())  # (B*S, K)
+            # broadcast-add in NHWC so the (B*S,K,H,W) result is
+            # channels_last without a transpose
+            yb = y_base.permute(0, 2, 3, 1).unsqueeze(1) \
+                + bias_pe.view(B, S, 1, 1, K).to(y_base.dtype)
+            return yb.view(B * S, y_base.shape[2], y_base.shape[3], K)
+
+        if use_side:
+            side.wait_stream(main)
+            with torch.cuda.stream(side):
+                yb = base_part()
+        else:
+            yb = base_part()
+
         if d:
             from mine_amd.ops.conv import conv3x3_reflect
             y_dec = conv3x3_reflect(x_dec, w[:, :d].contiguous(), None)
+            if use_side:
+                main.wait_stream(side)
+                yb.record_stream(main)
             yb = yb + y_dec.permute(0, 2, 3, 1)
+        elif use_side:
+            main.wait_stream(side)
+            yb.record_stream(main)
         return self.bn(yb.permute(0, 3, 1, 2))
 
 

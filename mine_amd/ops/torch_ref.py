@@ -299,3 +299,35 @@ def gather_pixel_by_pxpy(img: torch.Tensor, pxpy: torch.Tensor) -> torch.Tensor:
         py = torch.clamp(pp[:, 1:2], 0, H - 1)
         idx = px + W * py  # Bx1xN
     return torch.gather(img.reshape(B, C, H * W), 2, idx.expand(B, C, idx.size(2)))
+
+
+def get_xyz_from_depth(depth: torch.Tensor, K_inv: torch.Tensor) -> torch.Tensor:
+    """Backproject a depth map to camera-frame points: Bx1xHxW, Bx3x3 ->
+    Bx3xHxW (ref operations/mpi_rendering.py:85-105; unused in the
+    reference's train loop, kept for surface parity)."""
+    B, _, H, W = depth.shape
+    grid = make_meshgrid(H, W, device=depth.device)
+    rays = torch.matmul(K_inv, grid.reshape(3, -1).to(K_inv.dtype))  # Bx3xHW
+    return (rays * depth.reshape(B, 1, H * W)).reshape(B, 3, H, W)
+
+
+def disparity_consistency_src_to_tgt(src_disparity: torch.Tensor,
+                                     tgt_disparity: torch.Tensor,
+                                     G_tgt_src: torch.Tensor,
+                                     K_src_inv: torch.Tensor,
+                                     K_tgt: torch.Tensor) -> torch.Tensor:
+    """Cross-view disparity consistency: backproject the src disparity,
+    transform into the tgt camera, project, and compare the induced tgt
+    disparity with the rendered one at the projected pixels
+    (ref operations/mpi_rendering.py:108-137; unused in the reference's
+    train loop)."""
+    B, _, H, W = src_disparity.shape
+    xyz_src = get_xyz_from_depth(torch.reciprocal(src_disparity), K_src_inv)
+    xyz_tgt = tgt_plane_xyz(xyz_src.unsqueeze(1), G_tgt_src).squeeze(1)  # Bx3xHxW
+    z = xyz_tgt[:, 2:].clamp_min(1e-6)
+    pix = torch.matmul(K_tgt, xyz_tgt.reshape(B, 3, -1))
+    pxpy = pix[:, 0:2] / pix[:, 2:].clamp_min(1e-6)
+    sampled = gather_pixel_by_pxpy(tgt_disparity, pxpy)  # Bx1xHW
+    induced = torch.reciprocal(z.reshape(B, 1, -1))
+    return torch.mean(torch.abs(torch.log(sampled.clamp_min(1e-6))
+                                - torch.log(induced.clamp_min(1e-6))))

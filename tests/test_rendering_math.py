@@ -206,3 +206,26 @@ def test_alpha_composite_over():
     out2, w2 = tr.alpha_composite(a, val)
     torch.testing.assert_close(w2[:, 0], torch.full((B, 1, H, W), 0.5))
     torch.testing.assert_close(w2[:, 1], torch.full((B, 1, H, W), 0.25))
+
+
+def test_get_xyz_from_depth_and_consistency():
+    import torch
+    from mine_amd.ops import torch_ref as tr
+
+    B, H, W = 1, 8, 10
+    f = 6.0
+    K = torch.tensor([[f, 0, W / 2], [0, f, H / 2], [0, 0, 1.0]]).unsqueeze(0)
+    K_inv = torch.inverse(K)
+    depth = torch.full((B, 1, H, W), 4.0)
+    xyz = tr.get_xyz_from_depth(depth, K_inv)
+    assert xyz.shape == (B, 3, H, W)
+    torch.testing.assert_close(xyz[:, 2], depth[:, 0])
+    # center pixel backprojects onto the optical axis
+    torch.testing.assert_close(xyz[0, :2, H // 2, W // 2],
+                               torch.zeros(2), atol=1e-5, rtol=0)
+
+    # identity pose, identical constant disparity -> zero consistency loss
+    disp = torch.full((B, 1, H, W), 0.25)
+    G_id = torch.eye(4).unsqueeze(0)
+    loss = tr.disparity_consistency_src_to_tgt(disp, disp, G_id, K_inv, K)
+    assert float(loss) < 1e-6

@@ -1,5 +1,9 @@
 """Fused reflection-pad + 3x3 conv on the MFMA kernel.
 
+Carries the decoder's ConvBlock / Conv3x3 convolutions (the reference's
+ReflectionPad2d(1) + 3x3 nn.Conv2d, ref network/monodepth2/
+layers.py:106-138) on hand-written v_mfma_f32_16x16x32_bf16 tiles.
+
 Forward runs the hand-written v_mfma_f32_16x16x32_bf16 kernel
 (ops/csrc/conv_kernels.hip) with the pad folded into the LDS stage — no
 padded tensor is ever materialized. Backward materializes the pad once

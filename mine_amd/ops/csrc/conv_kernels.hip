@@ -1,4 +1,5 @@
 // Fused reflection-pad + 3x3 stride-1 conv, NHWC bf16, MFMA 16x16x32.
+// (the reference's ConvBlock conv, ref network/monodepth2/layers.py:106-138)
 //
 // The decoder's hot remaining convs are small-channel 3x3 blocks at
 // full resolution (16->16, 16->4, 32->16... at (B*S)=256 batch); MIOpen

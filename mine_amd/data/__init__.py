@@ -4,16 +4,15 @@ from mine_amd.data.synthetic import SyntheticMPIDataset, collate_src_tgt  # noqa
 def get_dataset(config, logger=None, is_validation: bool = False):
     """Dataset factory (ref train.py:69-103).
 
-    "llff" -> COLMAP-backed NeRFDataset; "synthetic" (and any benchmark
-    run without data on disk) -> SyntheticMPIDataset of the same item
-    schema. The remaining reference dataset names (realestate10k,
-    flowers, kitti_raw, dtu) use the synthetic generator when their
-    training_set_path does not exist — the reference never shipped those
-    pipelines either (ref train.py:100-101 raises NotImplementedError
-    for everything but llff).
+    "llff" -> COLMAP-backed NeRFDataset; "synthetic" ->
+    SyntheticMPIDataset of the same item schema. The remaining reference
+    dataset names (realestate10k, flowers, kitti_raw, dtu) raise
+    NotImplementedError exactly like the reference (ref train.py:100-101)
+    — a user pointing them at real data must not silently train on
+    noise. Benchmarks that want the synthetic generator under one of
+    those names opt in with ``data.allow_synthetic_fallback: true``
+    (bench.py builds SyntheticMPIDataset directly).
     """
-    import os
-
     name = config["data.name"]
     known = ("llff", "realestate10k", "flowers", "kitti_raw", "dtu", "synthetic")
     assert name in known, name
@@ -26,4 +25,14 @@ def get_dataset(config, logger=None, is_validation: bool = False):
                            supervision_count=config["data.num_tgt_views"],
                            visible_points_count=config["data.visible_point_count"],
                            img_pre_downsample_ratio=config["data.img_pre_downsample_ratio"])
+    if name != "synthetic" and not bool(
+            config.get("data.allow_synthetic_fallback", False)):
+        raise NotImplementedError(
+            f"dataset pipeline '{name}' is not implemented (the reference "
+            "raises here too, ref train.py:100-101); set data.name: "
+            "synthetic or data.allow_synthetic_fallback: true to train on "
+            "the synthetic generator")
+    if logger is not None and name != "synthetic":
+        logger.warning("dataset '%s': using the SYNTHETIC generator "
+                       "(data.allow_synthetic_fallback is set)", name)
     return SyntheticMPIDataset(config, is_validation=is_validation)

@@ -123,7 +123,11 @@ class SynthesisTask:
              {"params": self.decoder.parameters(), "lr": config["lr.decoder_lr"]}],
             weight_decay=config["lr.weight_decay"])
 
-        # rank-0 restore BEFORE the parameter broadcast (ref CS5).
+        # Restore BEFORE the parameter broadcast (ref CS5). Every rank
+        # that can see the checkpoint file loads it (workspace is a shared
+        # path on one node) — restoring on rank 0 only left the other
+        # ranks with cold Adam moments and no resume meta, desyncing
+        # epoch counts/LR schedules across ranks (round-1 ADVICE.md).
         # "auto" resumes from the workspace's checkpoint_latest.pth when
         # one exists — an elastic-restart convenience the reference
         # lacked (it always restarted at epoch 1).
@@ -133,10 +137,16 @@ class SynthesisTask:
                                 "checkpoint_latest.pth")
             ckpt_path = cand if os.path.exists(cand) else None
         self._restored_meta = {}
-        if self.state.is_rank0:
+        restored_here = False
+        if ckpt_path and (self.state.is_rank0 or os.path.exists(ckpt_path)):
+            # non-rank0 ranks skip silently if the file is unreachable
+            # (rank 0 keeps the reference's hard assert); they then
+            # receive the state by broadcast below.
             self._restored_meta = restore_model(
                 ckpt_path, self.backbone, self.decoder, self.optimizer,
-                logger=logger) or {}
+                logger=logger if self.state.is_rank0 else None) or {}
+            restored_here = True
+        self._sync_restored_state(restored_here)
 
         self.grad_engine = None
         if not is_val:
@@ -193,8 +203,45 @@ class SynthesisTask:
 
         self.current_epoch = 0
         self.global_step = 0
+        # device-side NaN-guard skip counter (read on logging steps only)
+        self._nan_skip_count: Optional[torch.Tensor] = None
 
     # ------------------------------------------------------------------
+    def _sync_restored_state(self, restored_here: bool) -> None:
+        """Make resume state rank-coherent (round-1 ADVICE.md items 1-2):
+        every rank must agree on the resume meta (epoch / global_step /
+        LR fast-forward — a mismatch desyncs per-rank collective counts),
+        and ranks that could not read the checkpoint file receive the
+        Adam state from rank 0 (parameters are broadcast separately by
+        GradAllReduceEngine at construction)."""
+        if self.is_val or not torch.distributed.is_initialized() or \
+                torch.distributed.get_world_size() <= 1:
+            return
+        dist = torch.distributed
+        flags = [None] * dist.get_world_size()
+        dist.all_gather_object(flags, bool(restored_here))
+        payload = {"meta": self._restored_meta}
+        if any(flags) and not all(flags) and self.state.is_rank0:
+            # ship Adam state to the ranks that missed the file
+            opt_sd = self.optimizer.state_dict()
+
+            def to_cpu(x):
+                if torch.is_tensor(x):
+                    return x.cpu()
+                if isinstance(x, dict):
+                    return {k: to_cpu(v) for k, v in x.items()}
+                if isinstance(x, list):
+                    return [to_cpu(v) for v in x]
+                return x
+            payload["optimizer"] = to_cpu(opt_sd)
+        obj = [payload if self.state.is_rank0 else None]
+        dist.broadcast_object_list(obj, src=0)
+        if not self.state.is_rank0:
+            self._restored_meta = obj[0].get("meta") or {}
+            opt_sd = obj[0].get("optimizer")
+            if opt_sd is not None and not restored_here:
+                self.optimizer.load_state_dict(opt_sd)
+
     def _autocast(self):
         if self.is_gpu and self.amp_dtype is not None:
             return torch.autocast(device_type="cuda", dtype=self.amp_dtype)
@@ -451,39 +498,14 @@ class SynthesisTask:
         mark("set_data")
         loss_dict, _ = self.loss_fcn(is_val=False)
         mark("forward")
-        # NaN guard (absent in the reference — SURVEY section 5c): a
-        # non-finite loss skips the update instead of poisoning the
-        # parameters. The skip decision is agreed ACROSS ranks (min over
-        # the finite flags) so every rank steps or skips together and the
-        # gradient buckets are still reduced — collectives stay matched.
-        if bool(self.config.get("training.nan_guard", True)):
-            finite = torch.isfinite(loss_dict["loss"].detach()).to(torch.int32)
-            if torch.distributed.is_initialized() and \
-                    torch.distributed.get_world_size() > 1:
-                finite = finite.clone()
-                torch.distributed.all_reduce(
-                    finite, op=torch.distributed.ReduceOp.MIN)
-            if not bool(finite.item()):
-                if self.logger:
-                    self.logger.warning(
-                        "non-finite loss at step %d; skipping update",
-                        self.global_step)
-                self._nan_skips = getattr(self, "_nan_skips", 0) + 1
-                if self.grad_engine is not None:
-                    # no backward: finish_step() reduces the zeroed buckets
-                    # itself, so the collectives stay matched across ranks
-                    self.grad_engine.zero_grad()
-                    self.grad_engine.finish_step()
-                else:
-                    self.optimizer.zero_grad(set_to_none=False)
-                mark("backward")
-                mark("optimizer")
-                return loss_dict
         if self.grad_engine is not None:
             self.grad_engine.zero_grad()
         else:
             self.optimizer.zero_grad(set_to_none=False)
         if self.grad_scaler is not None:
+            # fp16: the GradScaler already skips non-finite steps, and it
+            # checks the REDUCED gradients (finish_step runs before
+            # scaler.step), so the skip decision is rank-coherent.
             self.grad_scaler.scale(loss_dict["loss"]).backward()
             if self.grad_engine is not None:
                 self.grad_engine.finish_step()
@@ -495,6 +517,40 @@ class SynthesisTask:
             if self.grad_engine is not None:
                 self.grad_engine.finish_step()
             mark("backward")
+            # NaN guard (absent in the reference — SURVEY section 5c),
+            # fully DEVICE-SIDE: a non-finite loss zeroes the gradients so
+            # nothing poisons the parameters. No `.item()` host sync in
+            # the hot loop (the round-1 version stalled the pipeline every
+            # step — ADVICE.md item 3); the skip count is a device counter
+            # read only on (already-syncing) logging steps. The finite
+            # flag is all-reduced with MIN so every rank gates alike and
+            # the optimizer states stay bit-identical across ranks.
+            # (Adam still applies its momentum-decay update on a gated
+            # step — the parameters move slightly along stale momentum but
+            # no NaN enters; a pure skip would need a host sync.)
+            if bool(self.config.get("training.nan_guard", True)):
+                finite = torch.isfinite(
+                    loss_dict["loss"].detach()).to(torch.float32)
+                if torch.distributed.is_initialized() and \
+                        torch.distributed.get_world_size() > 1:
+                    torch.distributed.all_reduce(
+                        finite, op=torch.distributed.ReduceOp.MIN)
+                if self._nan_skip_count is None or \
+                        self._nan_skip_count.device != finite.device:
+                    self._nan_skip_count = torch.zeros_like(finite)
+                self._nan_skip_count += 1.0 - finite
+                if self.grad_engine is not None:
+                    grads = [b.flat for b in self.grad_engine.buckets]
+                else:
+                    grads = [p.grad for g in self.optimizer.param_groups
+                             for p in g["params"] if p.grad is not None]
+                # scale finite grads by the flag, then map NaN/inf -> 0
+                # (0 * NaN is NaN, so the multiply alone cannot zero a
+                # poisoned buffer); on healthy steps nan_to_num_ is a
+                # no-op pass over the ~6 bucket flats (~40 us at HBM bw)
+                torch._foreach_mul_(grads, finite)
+                for g in grads:
+                    torch.nan_to_num_(g, nan=0.0, posinf=0.0, neginf=0.0)
             self.optimizer.step()
         mark("optimizer")
         return loss_dict
@@ -543,9 +599,15 @@ class SynthesisTask:
             if step % ckpt_every == 0 and self.state.is_rank0 and \
                     self.state.local_workspace:
                 path = os.path.join(self.state.local_workspace, "checkpoint_latest.pth")
+                # meta["epoch"] records the last COMPLETED epoch: a
+                # mid-epoch save records epoch-1, so resume re-runs the
+                # interrupted epoch instead of silently skipping its
+                # remainder (round-1 ADVICE.md item 4). global_step keeps
+                # its mid-epoch value (monotone; re-run steps re-count).
                 save_checkpoint(path, self.backbone, self.decoder, self.optimizer,
-                                meta={"epoch": self.current_epoch,
-                                      "global_step": self.global_step})
+                                meta={"epoch": self.current_epoch - 1,
+                                      "global_step": self.global_step,
+                                      "mid_epoch": True})
                 if self.logger:
                     self.logger.info(f"Latest checkpoint saved at {path}")
 
@@ -571,6 +633,14 @@ class SynthesisTask:
             self.current_epoch = epoch
             self.train_epoch(train_loader, val_loader, epoch)
             self.lr_scheduler.step()
+            if self.state.is_rank0 and self.state.local_workspace:
+                # end-of-epoch latest save: meta marks the epoch COMPLETE
+                path = os.path.join(self.state.local_workspace,
+                                    "checkpoint_latest.pth")
+                save_checkpoint(path, self.backbone, self.decoder,
+                                self.optimizer,
+                                meta={"epoch": epoch,
+                                      "global_step": self.global_step})
             if self.state.is_rank0 and self.logger:
                 self.logger.info("Epoch finished, average losses: ")
                 for v in self.train_losses.values():
@@ -606,6 +676,14 @@ class SynthesisTask:
 
     # ------------------------------------------------------------------
     def _log_training(self, epoch, step, n_steps, loss_dict) -> None:
+        # logging steps already host-sync on float(loss): piggyback the
+        # NaN-guard skip counter read here (never in the hot loop)
+        if self._nan_skip_count is not None:
+            skips = int(self._nan_skip_count.item())
+            if skips and self.logger:
+                self.logger.warning(
+                    "NaN guard gated %d non-finite step(s) so far "
+                    "(gradients zeroed device-side)", skips)
         for key, meter in self.train_losses.items():
             v = float(loss_dict[key])
             meter.update(v)

@@ -237,6 +237,13 @@ def homography_grid_sample(src: torch.Tensor, H_src_tgt: torch.Tensor,
     gx = (uv[..., 0] + 0.5) / (W_src * 0.5) - 1.0
     gy = (uv[..., 1] + 0.5) / (H_src * 0.5) - 1.0
     sample_grid = torch.stack((gx, gy), dim=-1)
+    # grid_sample's backward derives integer scatter indices from the grid
+    # values WITHOUT an is-finite check — a NaN coordinate (poisoned batch,
+    # degenerate homography) segfaults the CPU kernel. Map non-finite
+    # coords far out of bounds: border padding clamps them in forward and
+    # the backward treats them as clipped (zero-gradient) taps.
+    sample_grid = torch.nan_to_num(sample_grid, nan=-10.0, posinf=-10.0,
+                                   neginf=-10.0)
     warped = F.grid_sample(src, sample_grid, mode="bilinear",
                            padding_mode="border", align_corners=False)
     return warped, valid

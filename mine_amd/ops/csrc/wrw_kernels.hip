@@ -1,7 +1,6 @@
-// EXPERIMENTAL (round-2 groundwork; compile-verified, NOT wired into the
-// training path — see docs/NEXT.md #1 and mine_amd/ops/conv.py's gate).
-//
-// Split-K MFMA weight-gradient (wrw) for the fused reflect-pad 3x3 conv:
+// Split-K MFMA weight-gradient (wrw) for the fused reflect-pad 3x3 conv
+// (wired into _Conv3x3ReflFn.backward in round 2; numerics tests
+// tests/test_gpu_ops.py::test_wrw_matches_torch):
 //   dW[k, c, dy, dx] = sum_{n,y,x} gy[n,y,x,k] * xpad[n, y+dy, x+dx, c]
 // as a GEMM with M = K (out-channels), N = 9*C taps, contraction over
 // the 25M pixels, on v_mfma_f32_16x16x32_bf16 (fragment maps per
@@ -15,8 +14,9 @@
 // flushes ONCE with fp32 atomics (512 slabs x K*9C cells — ~500 adds
 // per cell, negligible contention).
 //
-// Supported: C in {8..32}, K <= 32 (the decoder's full/half-res blocks),
-// W <= 1022, stride 1, pad 1 reflect.
+// Supported: C % 8 == 0 (grid.z splits the tap axis in 32-channel groups,
+// so C > 32 costs one extra gy stage per group), any K (grid.y chunks of
+// 16), W <= 1022, stride 1, pad 1 reflect.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -44,14 +44,17 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
                    float* __restrict__ dw,                 // (K, 9*C) packed
                    int N, int H, int W, int C, int K, int n_slabs) {
   extern __shared__ __hip_bfloat16 lds[];
-  // LDS layout: gyT [16][Wpad] then xT [C][3][W+2 pad8]
+  // LDS layout: gyT [16][Wpad] then xT [Cg][3][W+2 pad8] for this
+  // block's 32-channel group (grid.z picks the group; C <= 32 -> one)
   const int Wg = (W + 7) & ~7;          // gy row padded to 8
   const int Wx = (W + 2 + 7) & ~7;      // x row (+halo) padded to 8
   __hip_bfloat16* s_gy = lds;                       // 16 * Wg
-  __hip_bfloat16* s_x = lds + 16 * Wg;              // C * 3 * Wx
+  __hip_bfloat16* s_x = lds + 16 * Wg;              // Cg * 3 * Wx
 
   const int kc = blockIdx.y;            // k-chunk (16 out-channels)
   const int k0 = kc * 16;
+  const int c0 = blockIdx.z * MAX_C;    // channel-group base
+  const int Cg = (C - c0) < MAX_C ? (C - c0) : MAX_C;  // channels here
   const int slab = blockIdx.x;
   const int64_t rows_total = (int64_t)N * H;
   const int64_t r_begin = rows_total * slab / n_slabs;
@@ -59,8 +62,8 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int Cv = C / 8;
-  const int nch = (9 * Cv + 1) / 2;     // 16-wide N chunks over 9C
+  const int Cv = Cg / 8;
+  const int nch = (9 * Cv + 1) / 2;     // 16-wide N chunks over 9*Cg
 
   // accumulators: one (16k x 16taps) tile per n-chunk, 4 f32/lane
   f32x4 acc[MAX_NCH];
@@ -82,7 +85,7 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
           : (__hip_bfloat16)0.0f;
     }
     // ---- stage x^T rows y-1..y+1 reflected: s_x[c][row][xx] ----
-    for (int i = threadIdx.x; i < C * 3 * (W + 2); i += kBlock) {
+    for (int i = threadIdx.x; i < Cg * 3 * (W + 2); i += kBlock) {
       const int xx = i % (W + 2);
       const int rem = i / (W + 2);
       const int row = rem % 3;
@@ -90,7 +93,7 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
       const int yy = reflect1(y + row - 1, H);
       const int xs = reflect1(xx - 1, W);
       s_x[(c * 3 + row) * Wx + xx] =
-          x[(((int64_t)n * H + yy) * W + xs) * C + c];
+          x[(((int64_t)n * H + yy) * W + xs) * C + c0 + c];
     }
     __syncthreads();
 
@@ -119,7 +122,7 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
         //   the order col = (c*9 + dy*3 + dx)  [c-major taps]
         const int col = nc * 16 + (lane & 15);
         bf16x8 bfrag;
-        if (col < 9 * C) {
+        if (col < 9 * Cg) {
           const int c = col / 9;
           const int tap = col - c * 9;
           const int dy = tap / 3, dx = tap - dy * 3;
@@ -170,9 +173,10 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
       for (int i = lane; i < 256; i += 64) {
         const int krow = i / 16, jcol = i - (i / 16) * 16;
         const int kk = k0 + krow;
-        const int col = nc * 16 + jcol;
-        if (kk < K && col < 9 * C) {
-          atomicAdd(&dw[(int64_t)kk * 9 * C + col], red[i]);
+        const int col = nc * 16 + jcol;       // (c_local, tap) packed
+        if (kk < K && col < 9 * Cg) {
+          const int cl = col / 9, tap = col - cl * 9;
+          atomicAdd(&dw[((int64_t)kk * C + c0 + cl) * 9 + tap], red[i]);
         }
       }
     }
@@ -185,10 +189,11 @@ extern "C" void mine_conv3x3_wrw(const void* x, const void* gy, float* dw,
                                  int N, int H, int W, int C, int K,
                                  hipStream_t stream) {
   const int n_slabs = 512 < (int64_t)N * H ? 512 : (int)((int64_t)N * H);
+  const int Cg = C < MAX_C ? C : MAX_C;
   const int Wg = (W + 7) & ~7;
   const int Wx = (W + 2 + 7) & ~7;
-  const size_t lds = (16 * Wg + (size_t)C * 3 * Wx) * sizeof(__hip_bfloat16);
-  const dim3 grid(n_slabs, (K + 15) / 16);
+  const size_t lds = (16 * Wg + (size_t)Cg * 3 * Wx) * sizeof(__hip_bfloat16);
+  const dim3 grid(n_slabs, (K + 15) / 16, (C + MAX_C - 1) / MAX_C);
   hipLaunchKernelGGL(conv3x3_wrw_kernel, grid, dim3(kBlock), lds, stream,
                      reinterpret_cast<const __hip_bfloat16*>(x),
                      reinterpret_cast<const __hip_bfloat16*>(gy), dw,

@@ -467,12 +467,10 @@ def test_mfma_conv3x3_speed_vs_miopen():
     assert t_mfma < t_ref * 1.5  # must at least be in the same class
 
 
-@pytest.mark.skipif(os.environ.get("MINE_EXPERIMENTAL") != "1",
-                    reason="experimental kernel (round-2 groundwork); "
-                           "set MINE_EXPERIMENTAL=1 to run")
 @pytest.mark.parametrize("shape", [(3, 16, 12, 50, 16), (2, 32, 9, 64, 16),
-                                   (2, 16, 20, 70, 4)])
-def test_experimental_wrw_matches_torch(shape):
+                                   (2, 16, 20, 70, 4), (2, 64, 10, 48, 64),
+                                   (1, 40, 8, 52, 24)])
+def test_wrw_matches_torch(shape):
     import torch.nn.functional as F
     from mine_amd.ops.backend import get_extension
 
@@ -496,19 +494,19 @@ def test_experimental_wrw_matches_torch(shape):
     torch.testing.assert_close(dw.cpu(), w_probe.grad, rtol=2e-2, atol=2e-1)
 
 
-@pytest.mark.skipif(os.environ.get("MINE_EXPERIMENTAL") != "1",
-                    reason="experimental kernel (round-2 groundwork)")
-def test_experimental_bwd_data_matches_autograd():
+@pytest.mark.parametrize("B,C,H,W,K", [(2, 16, 10, 54, 16),
+                                       (2, 16, 12, 50, 4),
+                                       (1, 64, 9, 56, 32)])
+def test_bwd_data_matches_autograd(B, C, H, W, K):
     import torch.nn.functional as F
-    from mine_amd.ops.conv import conv3x3_bwd_data_experimental
+    from mine_amd.ops.conv import conv3x3_bwd_data
 
-    B, C, H, W, K = 2, 16, 10, 54, 16
     g = torch.Generator().manual_seed(23)
     x = torch.randn(B, C, H, W, generator=g)
     w = torch.randn(K, C, 3, 3, generator=g) * 0.2
     gy = torch.randn(B, K, H, W, generator=g)
 
-    gx = conv3x3_bwd_data_experimental(
+    gx = conv3x3_bwd_data(
         gy.to("cuda:0", torch.bfloat16).contiguous(
             memory_format=torch.channels_last),
         w.cuda())

@@ -151,8 +151,10 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
     # 4.2x vs pad+MIOpen at 256x16x256x384); for C>64 or narrow images
     # MIOpen's tuned igemm is better and the pad recompute in backward
     # is not paid back.
+    # (C % 16: the wrw kernel's c-groups are 16-wide; every decoder
+    # shape has C in {16, 32, 64})
     usable = (x.is_cuda and x.dtype == torch.bfloat16
-              and x.shape[1] % 8 == 0 and x.shape[1] <= 64
+              and x.shape[1] % 16 == 0 and x.shape[1] <= 64
               and x.shape[-1] >= 48 and x.shape[-2] >= 8
               and x.is_contiguous(memory_format=torch.channels_last))
     if usable:

@@ -242,7 +242,7 @@ at::Tensor conv3x3_wrw(at::Tensor x_flat, at::Tensor gy_flat, int64_t N,
               gy_flat.is_contiguous());
   TORCH_CHECK(x_flat.scalar_type() == at::kBFloat16 &&
               gy_flat.scalar_type() == at::kBFloat16);
-  TORCH_CHECK(C % 8 == 0 && W <= 1022);
+  TORCH_CHECK(C % 16 == 0 && W >= 2);
   auto dw = at::zeros({K, C, 3, 3}, x_flat.options().dtype(at::kFloat));
   mine_conv3x3_wrw(x_flat.data_ptr(), gy_flat.data_ptr(),
                    dw.data_ptr<float>(), (int)N, (int)H, (int)W, (int)C,

@@ -1,22 +1,30 @@
-// Split-K MFMA weight-gradient (wrw) for the fused reflect-pad 3x3 conv
-// (wired into _Conv3x3ReflFn.backward in round 2; numerics tests
-// tests/test_gpu_ops.py::test_wrw_matches_torch):
+// Split-K MFMA weight-gradient (wrw) for the fused reflect-pad 3x3 conv:
 //   dW[k, c, dy, dx] = sum_{n,y,x} gy[n,y,x,k] * xpad[n, y+dy, x+dx, c]
 // as a GEMM with M = K (out-channels), N = 9*C taps, contraction over
-// the 25M pixels, on v_mfma_f32_16x16x32_bf16 (fragment maps per
-// tools/mfma_probe.hip: A row=lane&15 k=(lane>>4)*8+e consecutive).
+// the pixels, on v_mfma_f32_16x16x32_bf16.
 //
-// Decomposition: grid.x = pixel slabs (few hundred), grid.y = K/16
-// k-chunks. Each workgroup walks its slab's rows, stages per row:
-//   LDS gy^T  [16][W]      (A operand: 8 consecutive PIXELS per lane)
-//   LDS x^T   [C][3][W+2]  (B operand: 8 consecutive x at (c, dy row))
-// keeps the (16 x 9C) partial in registers across the whole slab, and
-// flushes ONCE with fp32 atomics (512 slabs x K*9C cells — ~500 adds
-// per cell, negligible contention).
+// v2 design (the round-2 rewrite; v1 measured 44 ms/step at the
+// flagship config — strided 2-byte global staging and per-element LDS
+// fragment loads): both operands are staged with COALESCED bf16x8
+// loads + single b128 LDS writes into [4px][16ch] transpose-read
+// blocks, and every MFMA fragment is read with TWO
+// ds_read_b64_tr_b16 hardware transpose reads (lane l, elem j reads
+// lds[(l&15) + j*16 + (l>>4)*64] — the conflict-free pattern of the
+// CDNA4 guide). The 3 dx taps are handled by staging THREE
+// pixel-shifted copies of the gy tile (A operand), so every fragment
+// read stays 4-pixel-block aligned; the 3 dy taps are three staged x
+// rows (B operand), no duplication.
 //
-// Supported: C % 8 == 0 (grid.z splits the tap axis in 32-channel groups,
-// so C > 32 costs one extra gy stage per group), any K (grid.y chunks of
-// 16), W <= 1022, stride 1, pad 1 reflect.
+// Per j-chunk each wave issues 6 A transpose-reads + 12 B
+// transpose-reads + 18 MFMAs (Cg = 32) — MFMA-bound, vs v1's 144
+// scalar LDS reads per 18 MFMAs.
+//
+// Decomposition: grid = (pixel slabs, K/16 chunks, C/32 groups); a
+// block walks its slab's rows in 254-pixel tiles, accumulates the
+// (16k x 9*Cg) partial in registers across the whole slab, and
+// flushes ONCE with fp32 atomics.
+//
+// Supported: C % 16 == 0, any K, any W >= 2, stride 1, pad 1 reflect.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -26,11 +34,17 @@ namespace {
 
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
+using s16x4 = __attribute__((ext_vector_type(4))) short;
+using s16x8 = __attribute__((ext_vector_type(8))) short;
+using lds_short = __attribute__((address_space(3))) short;
+using lds_s16x4 = __attribute__((address_space(3))) s16x4;
 
 constexpr int kBlock = 256;
-constexpr int MAX_W = 1022;
-constexpr int MAX_C = 32;
-constexpr int MAX_NCH = 18;  // 9*C/16 <= 18 for C <= 32
+constexpr int P_TILE = 254;          // output pixels per x-tile
+constexpr int JP = 256;              // contraction span (P_TILE + 2)
+constexpr int NB = JP / 4;           // 4-pixel blocks per image row span
+constexpr int IMG_ELEMS = 16 * JP;   // one [16ch][JP px] image (4096)
+constexpr int N_JCHUNK = JP / 32;    // 8 contraction chunks of 32
 
 __device__ __forceinline__ int reflect1(int v, int n) {
   if (v < 0) v = -v;
@@ -38,145 +52,177 @@ __device__ __forceinline__ int reflect1(int v, int n) {
   return v;
 }
 
+// element offset of (col, j) inside one transpose-read image.
+// Blocks of [4 px][16 ch]; pixel-block p stored at index
+// (p&1)*NB/2 + p/2 so that a fragment's two tr reads (elems 0..3 and
+// 4..7 of lane group g = px 8g..8g+7) land at base and base + JP*8.
+__device__ __forceinline__ int img_elem(int col, int j) {
+  const int p = j >> 2;
+  const int bi = (p & 1) * (NB / 2) + (p >> 1);
+  return bi * 64 + (j & 3) * 16 + col;
+}
+
+__device__ __forceinline__ bf16x8 frag2(const __hip_bfloat16* lds_base,
+                                        int elem_off) {
+  const lds_short* p =
+      (const lds_short*)(lds_base) + elem_off;
+  s16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_s16x4*)(p));
+  s16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (lds_s16x4*)(p + JP * 8));
+  s16x8 s = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+  return __builtin_bit_cast(bf16x8, s);
+}
+
 __global__ void __launch_bounds__(kBlock)
 conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
                    const __hip_bfloat16* __restrict__ gy,  // (N,H,W,K)
-                   float* __restrict__ dw,                 // (K, 9*C) packed
+                   float* __restrict__ dw,                 // (K,C,3,3)
                    int N, int H, int W, int C, int K, int n_slabs) {
-  extern __shared__ __hip_bfloat16 lds[];
-  // LDS layout: gyT [16][Wpad] then xT [Cg][3][W+2 pad8] for this
-  // block's 32-channel group (grid.z picks the group; C <= 32 -> one)
-  const int Wg = (W + 7) & ~7;          // gy row padded to 8
-  const int Wx = (W + 2 + 7) & ~7;      // x row (+halo) padded to 8
-  __hip_bfloat16* s_gy = lds;                       // 16 * Wg
-  __hip_bfloat16* s_x = lds + 16 * Wg;              // Cg * 3 * Wx
+  // LDS: 3 A images (gy, dx-shifted) + 3*chalves B images (x rows)
+  extern __shared__ __attribute__((aligned(16))) __hip_bfloat16 lds[];
 
-  const int kc = blockIdx.y;            // k-chunk (16 out-channels)
-  const int k0 = kc * 16;
-  const int c0 = blockIdx.z * MAX_C;    // channel-group base
-  const int Cg = (C - c0) < MAX_C ? (C - c0) : MAX_C;  // channels here
-  const int slab = blockIdx.x;
+  const int k0 = blockIdx.y * 16;
+  const int c0 = blockIdx.z * 32;
+  const int Cg = (C - c0) < 32 ? (C - c0) : 32;   // 16 or 32
+  const int chalves = Cg / 16;
+  const int slab = (int)blockIdx.x;
   const int64_t rows_total = (int64_t)N * H;
   const int64_t r_begin = rows_total * slab / n_slabs;
   const int64_t r_end = rows_total * (slab + 1) / n_slabs;
 
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  const int Cv = Cg / 8;
-  const int nch = (9 * Cv + 1) / 2;     // 16-wide N chunks over 9*Cg
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
 
-  // accumulators: one (16k x 16taps) tile per n-chunk, 4 f32/lane
-  f32x4 acc[MAX_NCH];
+  __hip_bfloat16* s_A = lds;                       // [3][IMG_ELEMS]
+  __hip_bfloat16* s_B = lds + 3 * IMG_ELEMS;       // [3][chalves][IMG_ELEMS]
+
+  const int n_acc = 9 * chalves;                   // (dx, dy, ch)
+  f32x4 acc[18];
 #pragma unroll
-  for (int i = 0; i < MAX_NCH; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < 18; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int px_chunks = (W + 31) / 32;  // 32-pixel contraction chunks
+  const int kocts = 2;                              // 16 k = 2 bf16x8
+  const int cocts = 2 * chalves;                    // Cg = cocts * 8
 
   for (int64_t r = r_begin; r < r_end; ++r) {
     const int n = (int)(r / H);
     const int y = (int)(r % H);
+    const int64_t gy_row = (((int64_t)n * H + y) * W) * K;
 
-    // ---- stage gy^T: s_gy[k][xx] = gy[n,y,xx,k0+k] ----
-    for (int i = threadIdx.x; i < 16 * W; i += kBlock) {
-      const int k = i / W, xx = i - (i / W) * W;
-      const int kk = k0 + k;
-      s_gy[k * Wg + xx] = (kk < K)
-          ? gy[(((int64_t)n * H + y) * W + xx) * K + kk]
-          : (__hip_bfloat16)0.0f;
-    }
-    // ---- stage x^T rows y-1..y+1 reflected: s_x[c][row][xx] ----
-    for (int i = threadIdx.x; i < Cg * 3 * (W + 2); i += kBlock) {
-      const int xx = i % (W + 2);
-      const int rem = i / (W + 2);
-      const int row = rem % 3;
-      const int c = rem / 3;
-      const int yy = reflect1(y + row - 1, H);
-      const int xs = reflect1(xx - 1, W);
-      s_x[(c * 3 + row) * Wx + xx] =
-          x[(((int64_t)n * H + yy) * W + xs) * C + c0 + c];
-    }
-    __syncthreads();
+    for (int x0 = 0; x0 < W; x0 += P_TILE) {
+      const int Weff = (W - x0) < P_TILE ? (W - x0) : P_TILE;
 
-    // ---- contraction over this row's pixels, chunks of 32 ----
-    // waves split the chunks round-robin
-    for (int pc = wave; pc < px_chunks; pc += 4) {
-      const int p0 = pc * 32;
-      const int p_lane = p0 + (lane >> 4) * 8;  // this lane's 8 pixels
-      // A: gy^T row (lane&15) at 8 consecutive pixels; OOB pixels -> 0
-      bf16x8 afrag;
-      if (p_lane + 8 <= W) {
-        afrag = *reinterpret_cast<const bf16x8*>(
-            s_gy + (lane & 15) * Wg + p_lane);
-      } else {
+      // ---- zero the A images (covers dx edges + tile truncation) ----
+      __syncthreads();
+      {
+        s16x8* za = reinterpret_cast<s16x8*>(s_A);
+        for (int i = tid; i < 3 * IMG_ELEMS / 8; i += kBlock)
+          za[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+      __syncthreads();
+
+      // ---- stage A: gy row pixels, three dx-shifted copies ----------
+      for (int i = tid; i < P_TILE * kocts; i += kBlock) {
+        const int oct = i & 1;
+        const int pl = i >> 1;                      // pixel in tile
+        const int px = x0 + pl;
+        if (px >= W) continue;
+        const int kbase = k0 + oct * 8;
+        if (kbase >= K) continue;
+        s16x8 v;
+        if (kbase + 8 <= K) {
+          v = *reinterpret_cast<const s16x8*>(
+              reinterpret_cast<const short*>(gy + gy_row) + px * K + kbase);
+        } else {
+          const short* src =
+              reinterpret_cast<const short*>(gy + gy_row) + px * K;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int p = p_lane + e;
-          afrag[e] = (p < W) ? s_gy[(lane & 15) * Wg + p]
-                             : (__hip_bfloat16)0.0f;
+          for (int e = 0; e < 8; ++e)
+            v[e] = (kbase + e < K) ? src[kbase + e] : (short)0;
+        }
+#pragma unroll
+        for (int dx = 0; dx < 3; ++dx) {
+          const int j = pl + dx;
+          *reinterpret_cast<s16x8*>(
+              reinterpret_cast<short*>(s_A + dx * IMG_ELEMS) +
+              img_elem(oct * 8, j)) = v;
         }
       }
-      for (int nc = 0; nc < nch; ++nc) {
-        // B column (lane&15) of n-chunk nc -> tap index t = nc*16+(lane&15)
-        // over the (cb, tap, ci) k-ordering of the fwd kernel:
-        //   col j = (seg_lo..) ... here columns are (c, dy, dx) triples in
-        //   the order col = (c*9 + dy*3 + dx)  [c-major taps]
-        const int col = nc * 16 + (lane & 15);
-        bf16x8 bfrag;
-        if (col < 9 * Cg) {
-          const int c = col / 9;
-          const int tap = col - c * 9;
-          const int dy = tap / 3, dx = tap - dy * 3;
-          // xpad[y+dy-1, p+dx-1+1] = s_x[c][dy][p + dx]
-          const int base = (c * 3 + dy) * Wx + p_lane + dx;
-          if (p_lane + 8 <= W) {
-            // unaligned by dx: element loads (8x b16)
+
+      // ---- stage B: three reflected x rows, clamped halo ------------
+      for (int i = tid; i < 3 * JP * cocts; i += kBlock) {
+        const int coct = i % cocts;
+        const int rem = i / cocts;
+        const int j = rem % JP;
+        const int dy = rem / JP;
+        int u = x0 - 1 + j;
+        if (u > W) u = W;                           // clamp before reflect
+        const int us = reflect1(u, W);
+        const int yy = reflect1(y + dy - 1, H);
+        const s16x8 v = *reinterpret_cast<const s16x8*>(
+            reinterpret_cast<const short*>(x) +
+            ((((int64_t)n * H + yy) * W + us) * C + c0 + coct * 8));
+        *reinterpret_cast<s16x8*>(
+            reinterpret_cast<short*>(
+                s_B + (dy * chalves + (coct >> 1)) * IMG_ELEMS) +
+            img_elem((coct & 1) * 8, j)) = v;
+      }
+      __syncthreads();
+
+      // ---- contraction: j-chunks split across the 4 waves -----------
+      // A is nonzero only for j < Weff + 2
+      const int jchunks = (Weff + 2 + 31) / 32;
+      for (int jc = wave; jc < jchunks; jc += 4) {
+        const int ebase = jc * 32 * 8;              // j0 * 8 elems
+        bf16x8 afrag[3];
 #pragma unroll
-            for (int e = 0; e < 8; ++e) bfrag[e] = s_x[base + e];
-          } else {
+        for (int dx = 0; dx < 3; ++dx)
+          afrag[dx] = frag2(s_A + dx * IMG_ELEMS, ebase);
 #pragma unroll
-            for (int e = 0; e < 8; ++e) {
-              const int p = p_lane + e;
-              bfrag[e] = (p < W) ? s_x[base + e] : (__hip_bfloat16)0.0f;
+        for (int dy = 0; dy < 3; ++dy) {
+          for (int ch = 0; ch < chalves; ++ch) {
+            const bf16x8 bfrag =
+                frag2(s_B + (dy * chalves + ch) * IMG_ELEMS, ebase);
+#pragma unroll
+            for (int dx = 0; dx < 3; ++dx) {
+              const int ai = (dx * 3 + dy) * chalves + ch;
+              acc[ai] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  afrag[dx], bfrag, acc[ai], 0, 0, 0);
             }
           }
-        } else {
-#pragma unroll
-          for (int e = 0; e < 8; ++e) bfrag[e] = (__hip_bfloat16)0.0f;
         }
-        acc[nc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                          acc[nc], 0, 0, 0);
       }
     }
-    __syncthreads();
   }
 
-  // ---- flush: reduce the 4 wave-partials via LDS, one atomic per cell --
-  // (reuse the staging LDS as a fp32 scratch of 16 x 16 per n-chunk)
+  // ---- flush: reduce the 4 wave-partials via LDS, atomics to global --
   float* red = reinterpret_cast<float*>(lds);
-  for (int nc = 0; nc < nch; ++nc) {
+  for (int ai = 0; ai < n_acc; ++ai) {
+    const int dx = ai / (3 * chalves);
+    const int dy = (ai / chalves) % 3;
+    const int ch = ai % chalves;
     __syncthreads();
-    if (wave == 0) {
-      for (int i = threadIdx.x; i < 256; i += 64) red[i] = 0.0f;
-    }
+    if (wave == 0)
+      for (int i = tid; i < 256; i += 64) red[i] = 0.0f;
     __syncthreads();
     {
       const int jcol = lane & 15;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const int krow = (lane >> 4) * 4 + rr;
-        atomicAdd(&red[krow * 16 + jcol], acc[nc][rr]);
+        atomicAdd(&red[krow * 16 + jcol], acc[ai][rr]);
       }
     }
     __syncthreads();
-    // wave 0 writes the workgroup partial to global
     if (wave == 0) {
       for (int i = lane; i < 256; i += 64) {
-        const int krow = i / 16, jcol = i - (i / 16) * 16;
+        const int krow = i >> 4, jcol = i & 15;
         const int kk = k0 + krow;
-        const int col = nc * 16 + jcol;       // (c_local, tap) packed
-        if (kk < K && col < 9 * Cg) {
-          const int cl = col / 9, tap = col - cl * 9;
-          atomicAdd(&dw[((int64_t)kk * C + c0 + cl) * 9 + tap], red[i]);
+        const int cc = c0 + ch * 16 + jcol;
+        if (kk < K && cc < C) {
+          atomicAdd(&dw[(((int64_t)kk * C + cc) * 3 + dy) * 3 + dx],
+                    red[i]);
         }
       }
     }
@@ -188,14 +234,18 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
 extern "C" void mine_conv3x3_wrw(const void* x, const void* gy, float* dw,
                                  int N, int H, int W, int C, int K,
                                  hipStream_t stream) {
-  const int n_slabs = 512 < (int64_t)N * H ? 512 : (int)((int64_t)N * H);
-  const int Cg = C < MAX_C ? C : MAX_C;
-  const int Wg = (W + 7) & ~7;
-  const int Wx = (W + 2 + 7) & ~7;
-  const size_t lds = (16 * Wg + (size_t)Cg * 3 * Wx) * sizeof(__hip_bfloat16);
-  const dim3 grid(n_slabs, (K + 15) / 16, (C + MAX_C - 1) / MAX_C);
-  hipLaunchKernelGGL(conv3x3_wrw_kernel, grid, dim3(kBlock), lds, stream,
+  const int kc = (K + 15) / 16;
+  const int zc = (C + 31) / 32;
+  int slabs = 512 / (kc * zc);
+  if (slabs < 1) slabs = 1;
+  if ((int64_t)slabs > (int64_t)N * H) slabs = (int)((int64_t)N * H);
+  const int chalves = C >= 32 ? 2 : 1;
+  const size_t lds_bytes =
+      (size_t)(3 + 3 * chalves) * IMG_ELEMS * sizeof(__hip_bfloat16);
+  const dim3 grid(slabs, kc, zc);
+  hipLaunchKernelGGL(conv3x3_wrw_kernel, grid, dim3(kBlock), lds_bytes,
+                     stream,
                      reinterpret_cast<const __hip_bfloat16*>(x),
                      reinterpret_cast<const __hip_bfloat16*>(gy), dw,
-                     N, H, W, C, K, n_slabs);
+                     N, H, W, C, K, slabs);
 }

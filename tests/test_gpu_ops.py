@@ -469,7 +469,7 @@ def test_mfma_conv3x3_speed_vs_miopen():
 
 @pytest.mark.parametrize("shape", [(3, 16, 12, 50, 16), (2, 32, 9, 64, 16),
                                    (2, 16, 20, 70, 4), (2, 64, 10, 48, 64),
-                                   (1, 40, 8, 52, 24)])
+                                   (1, 48, 8, 52, 24), (1, 16, 3, 300, 16)])
 def test_wrw_matches_torch(shape):
     import torch.nn.functional as F
     from mine_amd.ops.backend import get_extension

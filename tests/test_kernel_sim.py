@@ -23,61 +23,109 @@ def _reflect1(v, n):
     return v
 
 
-def test_wrw_kernel_index_math():
-    """Simulates conv3x3_wrw_kernel: slab split, gy^T / x^T staging with
-    reflect halo, 32-pixel contraction chunks, (c*9+tap) column order."""
+def _img_elem(col, j, NB=64):
+    """ops/csrc/wrw_kernels.hip img_elem: the permuted-block
+    transpose-read image layout."""
+    p = j >> 2
+    bi = (p & 1) * (NB // 2) + (p >> 1)
+    return bi * 64 + (j & 3) * 16 + col
+
+
+def test_wrw_v2_tr_image_addressing():
+    """The transpose-read image layout is bijective and every
+    ds_read_b64_tr_b16 pair (base + (l&15) + jj*16 + (l>>4)*64, and the
+    same at +JP*8) reconstructs exactly fragment element
+    (col = l&15, j = j0 + (l>>4)*8 + e) — the MFMA A/B element map."""
+    JP = 256
+    img = {}
+    for col in range(16):
+        for j in range(JP):
+            off = _img_elem(col, j)
+            assert off not in img
+            img[off] = (col, j)
+    assert len(img) == 16 * JP and max(img) == 16 * JP - 1
+
+    for jc in range(JP // 32):
+        base = jc * 32 * 8
+        for lane in range(64):
+            g = lane >> 4
+            for jj in range(4):
+                off_lo = base + (lane & 15) + jj * 16 + g * 64
+                assert img[off_lo] == (lane & 15, jc * 32 + g * 8 + jj)
+                off_hi = off_lo + JP * 8
+                assert img[off_hi] == (lane & 15, jc * 32 + g * 8 + 4 + jj)
+
+
+def test_wrw_v2_kernel_index_math():
+    """Simulates the v2 conv3x3_wrw_kernel: slab split, x-tiles, three
+    dx-shifted A (gy) copies, clamped-reflect B (x) halo rows, 32-wide
+    j-chunks, (k, c, dy, dx) accumulator map."""
     torch.manual_seed(4)
-    B, C, H, W, K = 2, 8, 5, 40, 16
+    B, C, H, W, K = 2, 16, 5, 40, 24
+    P_TILE, JP = 254, 256
     x = torch.randn(B, C, H, W)
     gy = torch.randn(B, K, H, W)
 
-    Cv = C // 8
-    nch = (9 * Cv + 1) // 2
-    px_chunks = (W + 31) // 32
-    n_slabs = min(512, B * H)
     xn = x.permute(0, 2, 3, 1).double()
     gyn = gy.permute(0, 2, 3, 1).double()
-    dw = torch.zeros(K, 9 * C, dtype=torch.float64)
+    dw = torch.zeros(K, C, 3, 3, dtype=torch.float64)
+    n_slabs = min(4, B * H)
 
     for kc in range((K + 15) // 16):
         k0 = kc * 16
-        for slab in range(n_slabs):
-            r0, r1 = B * H * slab // n_slabs, B * H * (slab + 1) // n_slabs
-            part = torch.zeros(16, nch * 16, dtype=torch.float64)
-            for r in range(r0, r1):
-                n, y = divmod(r, H)
-                s_gy = torch.zeros(16, W, dtype=torch.float64)
-                for k in range(16):
-                    if k0 + k < K:
-                        s_gy[k] = gyn[n, y, :, k0 + k]
-                s_x = torch.zeros(C, 3, W + 2, dtype=torch.float64)
-                for c in range(C):
-                    for row in range(3):
-                        yy = _reflect1(y + row - 1, H)
-                        for xx in range(W + 2):
-                            s_x[c, row, xx] = xn[n, yy, _reflect1(xx - 1, W), c]
-                for pc in range(px_chunks):
-                    p0 = pc * 32
-                    A = torch.zeros(16, 32, dtype=torch.float64)
-                    Bm = torch.zeros(32, nch * 16, dtype=torch.float64)
-                    for p in range(32):
-                        if p0 + p < W:
-                            A[:, p] = s_gy[:, p0 + p]
-                    for col in range(nch * 16):
-                        if col < 9 * C:
-                            c, tap = divmod(col, 9)
-                            dy, dx = divmod(tap, 3)
-                            for p in range(32):
-                                if p0 + p < W:
-                                    Bm[p, col] = s_x[c, dy, p0 + p + dx]
-                    part += A @ Bm
-            dw[k0:k0 + min(16, K - k0)] += part[:min(16, K - k0), :9 * C]
-    dw = dw.view(K, C, 3, 3).float()
+        for zc in range((C + 31) // 32):
+            c0 = zc * 32
+            Cg = min(32, C - c0)
+            chalves = Cg // 16
+            for slab in range(n_slabs):
+                r0 = B * H * slab // n_slabs
+                r1 = B * H * (slab + 1) // n_slabs
+                acc = torch.zeros(3, 3, chalves, 16, 16, dtype=torch.float64)
+                for r in range(r0, r1):
+                    n, y = divmod(r, H)
+                    for x0 in range(0, W, P_TILE):
+                        Weff = min(P_TILE, W - x0)
+                        A = torch.zeros(3, 16, JP, dtype=torch.float64)
+                        for pl in range(P_TILE):
+                            px = x0 + pl
+                            if px >= W:
+                                continue
+                            for k in range(16):
+                                if k0 + k < K:
+                                    for dx in range(3):
+                                        A[dx, k, pl + dx] = gyn[n, y, px, k0 + k]
+                        Bm = torch.zeros(3, chalves, 16, JP,
+                                         dtype=torch.float64)
+                        for j in range(JP):
+                            u = min(x0 - 1 + j, W)
+                            us = _reflect1(u, W)
+                            for dy in range(3):
+                                yy = _reflect1(y + dy - 1, H)
+                                for ch in range(chalves):
+                                    for cc in range(16):
+                                        Bm[dy, ch, cc, j] = \
+                                            xn[n, yy, us, c0 + ch * 16 + cc]
+                        jchunks = (Weff + 2 + 31) // 32
+                        jmax = jchunks * 32
+                        for dx in range(3):
+                            for dy in range(3):
+                                for ch in range(chalves):
+                                    acc[dx, dy, ch] += \
+                                        A[dx, :, :jmax] @ Bm[dy, ch, :, :jmax].T
+                for dx in range(3):
+                    for dy in range(3):
+                        for ch in range(chalves):
+                            for krow in range(16):
+                                for jcol in range(16):
+                                    kk, cc = k0 + krow, c0 + ch * 16 + jcol
+                                    if kk < K and cc < C:
+                                        dw[kk, cc, dy, dx] += \
+                                            acc[dx, dy, ch, krow, jcol]
 
     w_probe = torch.zeros(K, C, 3, 3, requires_grad=True)
     y = F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"), w_probe)
     (y * gy).sum().backward()
-    torch.testing.assert_close(dw, w_probe.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(dw.float(), w_probe.grad, rtol=1e-4, atol=1e-4)
 
 
 def test_fwd_kernel_pack_and_zero_embed_math():

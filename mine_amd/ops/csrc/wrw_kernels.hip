@@ -64,8 +64,13 @@ __device__ __forceinline__ int img_elem(int col, int j) {
 
 __device__ __forceinline__ bf16x8 frag2(const __hip_bfloat16* lds_base,
                                         int elem_off) {
-  const lds_short* p =
-      (const lds_short*)(lds_base) + elem_off;
+  // ds_read_b64_tr_b16 has NO internal lane offset: each lane passes
+  // its own 8-B-aligned address covering its 4-element slice of the
+  // group's [4][16] block; the hardware redistributes column (l&15)
+  // of the block to lane l (CDNA4 guide §2 / T10).
+  const int l = threadIdx.x & 63;
+  const lds_short* p = (const lds_short*)(lds_base) + elem_off +
+                       (l & 15) * 4 + (l >> 4) * 64;
   s16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_s16x4*)(p));
   s16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
       (lds_s16x4*)(p + JP * 8));

@@ -78,18 +78,18 @@ __device__ __forceinline__ bf16x8 frag2(const __hip_bfloat16* lds_base,
   return __builtin_bit_cast(bf16x8, s);
 }
 
+template <int CHALVES>   // compile-time so acc[] indexing stays in regs
 __global__ void __launch_bounds__(kBlock)
 conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
                    const __hip_bfloat16* __restrict__ gy,  // (N,H,W,K)
                    float* __restrict__ dw,                 // (K,C,3,3)
                    int N, int H, int W, int C, int K, int n_slabs) {
-  // LDS: 3 A images (gy, dx-shifted) + 3*chalves B images (x rows)
+  // LDS: 3 A images (gy, dx-shifted) + 3*CHALVES B images (x rows)
   extern __shared__ __attribute__((aligned(16))) __hip_bfloat16 lds[];
 
   const int k0 = blockIdx.y * 16;
-  const int c0 = blockIdx.z * 32;
-  const int Cg = (C - c0) < 32 ? (C - c0) : 32;   // 16 or 32
-  const int chalves = Cg / 16;
+  const int c0 = blockIdx.z * CHALVES * 16;
+  constexpr int chalves = CHALVES;
   const int slab = (int)blockIdx.x;
   const int64_t rows_total = (int64_t)N * H;
   const int64_t r_begin = rows_total * slab / n_slabs;
@@ -102,13 +102,13 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
   __hip_bfloat16* s_A = lds;                       // [3][IMG_ELEMS]
   __hip_bfloat16* s_B = lds + 3 * IMG_ELEMS;       // [3][chalves][IMG_ELEMS]
 
-  const int n_acc = 9 * chalves;                   // (dx, dy, ch)
-  f32x4 acc[18];
+  constexpr int n_acc = 9 * CHALVES;               // (dx, dy, ch)
+  f32x4 acc[n_acc];
 #pragma unroll
-  for (int i = 0; i < 18; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < n_acc; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int kocts = 2;                              // 16 k = 2 bf16x8
-  const int cocts = 2 * chalves;                    // Cg = cocts * 8
+  constexpr int kocts = 2;                          // 16 k = 2 bf16x8
+  constexpr int cocts = 2 * CHALVES;                // Cg = cocts * 8
 
   for (int64_t r = r_begin; r < r_end; ++r) {
     const int n = (int)(r / H);
@@ -157,7 +157,7 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
 
       // ---- stage B: three reflected x rows, clamped halo ------------
       for (int i = tid; i < 3 * JP * cocts; i += kBlock) {
-        const int coct = i % cocts;
+        const int coct = i % cocts;           // cocts is constexpr
         const int rem = i / cocts;
         const int j = rem % JP;
         const int dy = rem / JP;
@@ -186,12 +186,13 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
           afrag[dx] = frag2(s_A + dx * IMG_ELEMS, ebase);
 #pragma unroll
         for (int dy = 0; dy < 3; ++dy) {
-          for (int ch = 0; ch < chalves; ++ch) {
+#pragma unroll
+          for (int ch = 0; ch < CHALVES; ++ch) {
             const bf16x8 bfrag =
-                frag2(s_B + (dy * chalves + ch) * IMG_ELEMS, ebase);
+                frag2(s_B + (dy * CHALVES + ch) * IMG_ELEMS, ebase);
 #pragma unroll
             for (int dx = 0; dx < 3; ++dx) {
-              const int ai = (dx * 3 + dy) * chalves + ch;
+              const int ai = (dx * 3 + dy) * CHALVES + ch;
               acc[ai] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                   afrag[dx], bfrag, acc[ai], 0, 0, 0);
             }
@@ -203,10 +204,11 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
 
   // ---- flush: reduce the 4 wave-partials via LDS, atomics to global --
   float* red = reinterpret_cast<float*>(lds);
+#pragma unroll
   for (int ai = 0; ai < n_acc; ++ai) {
-    const int dx = ai / (3 * chalves);
-    const int dy = (ai / chalves) % 3;
-    const int ch = ai % chalves;
+    const int dx = ai / (3 * CHALVES);
+    const int dy = (ai / CHALVES) % 3;
+    const int ch = ai % CHALVES;
     __syncthreads();
     if (wave == 0)
       for (int i = tid; i < 256; i += 64) red[i] = 0.0f;
@@ -239,18 +241,25 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
 extern "C" void mine_conv3x3_wrw(const void* x, const void* gy, float* dw,
                                  int N, int H, int W, int C, int K,
                                  hipStream_t stream) {
+  const int ch = (C % 32 == 0) ? 2 : 1;   // 32- or 16-channel groups
   const int kc = (K + 15) / 16;
-  const int zc = (C + 31) / 32;
+  const int zc = C / (16 * ch);
   int slabs = 512 / (kc * zc);
   if (slabs < 1) slabs = 1;
   if ((int64_t)slabs > (int64_t)N * H) slabs = (int)((int64_t)N * H);
-  const int chalves = C >= 32 ? 2 : 1;
   const size_t lds_bytes =
-      (size_t)(3 + 3 * chalves) * IMG_ELEMS * sizeof(__hip_bfloat16);
+      (size_t)(3 + 3 * ch) * IMG_ELEMS * sizeof(__hip_bfloat16);
   const dim3 grid(slabs, kc, zc);
-  hipLaunchKernelGGL(conv3x3_wrw_kernel, grid, dim3(kBlock), lds_bytes,
-                     stream,
-                     reinterpret_cast<const __hip_bfloat16*>(x),
-                     reinterpret_cast<const __hip_bfloat16*>(gy), dw,
-                     N, H, W, C, K, slabs);
+  if (ch == 2)
+    hipLaunchKernelGGL(conv3x3_wrw_kernel<2>, grid, dim3(kBlock), lds_bytes,
+                       stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x),
+                       reinterpret_cast<const __hip_bfloat16*>(gy), dw,
+                       N, H, W, C, K, slabs);
+  else
+    hipLaunchKernelGGL(conv3x3_wrw_kernel<1>, grid, dim3(kBlock), lds_bytes,
+                       stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x),
+                       reinterpret_cast<const __hip_bfloat16*>(gy), dw,
+                       N, H, W, C, K, slabs);
 }

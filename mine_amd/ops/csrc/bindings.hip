@@ -23,8 +23,8 @@ void mine_tgt_composite_fwd(const float*, const float*, const float*,
                             float*, int, int, int, int, int, hipStream_t);
 void mine_tgt_composite_bwd(const float*, const float*, const float*,
                             const float*, const float*, const float*,
-                            const float*, float*, int, int, int, int, int,
-                            hipStream_t);
+                            const float*, const float*, float*, float*,
+                            int, int, int, int, int, int, hipStream_t);
 void mine_reflect_pad_fwd_f32(const float*, float*, int, int, int, int, int,
                               hipStream_t);
 void mine_reflect_pad_fwd_bf16(const void*, void*, int, int, int, int, int,
@@ -149,17 +149,30 @@ std::vector<at::Tensor> tgt_composite_fwd(at::Tensor mpi, at::Tensor hinv,
   return {rgb, depth, mask};
 }
 
-at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor m,
-                             at::Tensor tvec, at::Tensor depths, bool bg_inf,
-                             at::Tensor g_rgb, at::Tensor g_depth) {
+// mode 1 (default): gather redesign — interior pixels write plain
+// per-plane payloads, a per-src-tile gather kernel inverts the map
+// (hfwd required). mode 0: the round-1 all-atomic scatter (kept for
+// A/B tests and as a fallback).
+at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor hfwd,
+                             at::Tensor m, at::Tensor tvec, at::Tensor depths,
+                             bool bg_inf, at::Tensor g_rgb, at::Tensor g_depth,
+                             int64_t mode) {
   CHECK_IN(mpi);
   const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
   auto grad_mpi = at::zeros_like(mpi);
+  at::Tensor payload;
+  float* pay_ptr = nullptr;
+  if (mode == 1) {
+    TORCH_CHECK(hfwd.numel() == (int64_t)B * S * 9, "hfwd required in gather mode");
+    payload = at::zeros_like(mpi);
+    pay_ptr = payload.data_ptr<float>();
+  }
   mine_tgt_composite_bwd(mpi.data_ptr<float>(), hinv.data_ptr<float>(),
+                         mode == 1 ? hfwd.data_ptr<float>() : nullptr,
                          m.data_ptr<float>(), tvec.data_ptr<float>(),
                          depths.data_ptr<float>(), optr(g_rgb), optr(g_depth),
-                         grad_mpi.data_ptr<float>(), B, S, H, W,
-                         bg_inf ? 1 : 0, stream());
+                         grad_mpi.data_ptr<float>(), pay_ptr, B, S, H, W,
+                         bg_inf ? 1 : 0, (int)mode, stream());
   return grad_mpi;
 }
 

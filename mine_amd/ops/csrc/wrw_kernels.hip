@@ -118,12 +118,29 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
     for (int x0 = 0; x0 < W; x0 += P_TILE) {
       const int Weff = (W - x0) < P_TILE ? (W - x0) : P_TILE;
 
+      // Only j < Weff + 2 is ever contracted (jchunks bound below), so
+      // staging beyond jstage is skipped — at W = 96 this saves the
+      // 2.6x full-JP staging waste. Rounded to the 32-wide chunk so
+      // every contracted element is initialized.
+      const int jstage = ((Weff + 2 + 31) / 32) * 32;
+
       // ---- zero the A images (covers dx edges + tile truncation) ----
       __syncthreads();
       {
-        s16x8* za = reinterpret_cast<s16x8*>(s_A);
-        for (int i = tid; i < 3 * IMG_ELEMS / 8; i += kBlock)
-          za[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        // blocks of 4 px: zero img blocks for j < jstage. The permuted
+        // block order (even blocks first) keeps the zeroed region two
+        // contiguous spans per image.
+        const int nb = jstage / 4;
+        const int span = (nb / 2 + (nb & 1)) * 64 / 8;  // even-half s16x8s
+        const int span2 = (nb / 2) * 64 / 8;            // odd half
+        s16x8 z{0, 0, 0, 0, 0, 0, 0, 0};
+        for (int d = 0; d < 3; ++d) {
+          s16x8* za = reinterpret_cast<s16x8*>(s_A + d * IMG_ELEMS);
+          for (int i = tid; i < span; i += kBlock) za[i] = z;
+          s16x8* zb = reinterpret_cast<s16x8*>(s_A + d * IMG_ELEMS) +
+                      (NB / 2) * 64 / 8;
+          for (int i = tid; i < span2; i += kBlock) zb[i] = z;
+        }
       }
       __syncthreads();
 
@@ -161,6 +178,7 @@ conv3x3_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,H,W,C)
         const int rem = i / cocts;
         const int j = rem % JP;
         const int dy = rem / JP;
+        if (j >= jstage) continue;
         int u = x0 - 1 + j;
         if (u > W) u = W;                           // clamp before reflect
         const int us = reflect1(u, W);

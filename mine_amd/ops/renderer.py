@@ -32,6 +32,19 @@ import torch
 from mine_amd.ops import torch_ref as tr
 from mine_amd.ops.backend import get_extension
 
+_TGT_BWD_MODE = None
+
+
+def tgt_bwd_mode() -> int:
+    """1 (default) = gather-based warp backward (per-src-tile inversion,
+    no interior atomics); 0 = the round-1 all-atomic bilinear scatter.
+    Override with MINE_TGT_BWD=scatter."""
+    global _TGT_BWD_MODE
+    if _TGT_BWD_MODE is None:
+        import os
+        _TGT_BWD_MODE = 0 if os.environ.get("MINE_TGT_BWD") == "scatter" else 1
+    return _TGT_BWD_MODE
+
 
 # ---------------------------------------------------------------------------
 # packing helpers
@@ -166,10 +179,22 @@ class _TgtCompositeFn(torch.autograd.Function):
         mpi, hinv, m, tvec, depths = ctx.saved_tensors
         ext = get_extension(required=True)
         empty = torch.empty(0, device=mpi.device, dtype=torch.float32)
+        mode = tgt_bwd_mode()
+        if mode == 1:
+            # gather redesign: the per-src-tile inversion needs the
+            # forward (src -> tgt pixel) homographies
+            from mine_amd.utils.geometry import inverse_3x3
+            with torch.no_grad():
+                B, S = hinv.shape[0], hinv.shape[1]
+                hfwd = inverse_3x3(hinv.reshape(B * S, 3, 3)) \
+                    .reshape(B, S, 3, 3).contiguous()
+        else:
+            hfwd = empty
         grad_mpi = ext.tgt_composite_bwd(
-            mpi, hinv, m, tvec, depths, ctx.bg_inf,
+            mpi, hinv, hfwd, m, tvec, depths, ctx.bg_inf,
             g_rgb.contiguous() if g_rgb is not None else empty,
-            g_depth.contiguous() if g_depth is not None else empty)
+            g_depth.contiguous() if g_depth is not None else empty,
+            mode)
         return grad_mpi, None, None, None, None, None
 
 

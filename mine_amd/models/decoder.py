@@ -198,14 +198,17 @@ class MPIDecoder(nn.Module):
         })
 
     def _up(self, x: torch.Tensor) -> torch.Tensor:
-        """Nearest x2 upsample, autocast-shielded: CUDA autocast promotes
-        upsample_nearest to fp32, which poisoned the downstream convs/pads
-        into fp32 (observed in profiles); outside autocast the op keeps
-        the tensor's own dtype."""
+        """Nearest x2 upsample on the HIP stream kernel (torch's
+        channels_last nearest kernel measured ~15x off roofline),
+        autocast-shielded: CUDA autocast promotes upsample_nearest to
+        fp32, which poisoned the downstream convs/pads into fp32
+        (observed in profiles); outside autocast the op keeps the
+        tensor's own dtype."""
+        from mine_amd.ops.upsample import upsample_nearest2x
         if x.is_cuda and torch.is_autocast_enabled():
             with torch.autocast("cuda", enabled=False):
-                return self.upsample(x)
-        return self.upsample(x)
+                return upsample_nearest2x(x)
+        return upsample_nearest2x(x)
 
     def _expand_with_pe(self, feat: torch.Tensor, pe: torch.Tensor,
                         B: int, S: int) -> torch.Tensor:

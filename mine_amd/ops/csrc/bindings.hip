@@ -25,6 +25,10 @@ void mine_tgt_composite_bwd(const float*, const float*, const float*,
                             const float*, const float*, const float*,
                             const float*, const float*, float*, float*,
                             int, int, int, int, int, int, hipStream_t);
+void mine_upsample2x_fwd(const void*, void*, int64_t, int64_t, int64_t,
+                         int64_t, int, hipStream_t);
+void mine_upsample2x_bwd(const void*, void*, int64_t, int64_t, int64_t,
+                         int64_t, int, hipStream_t);
 void mine_reflect_pad_fwd_f32(const float*, float*, int, int, int, int, int,
                               hipStream_t);
 void mine_reflect_pad_fwd_bf16(const void*, void*, int, int, int, int, int,
@@ -174,6 +178,34 @@ at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor hfwd,
                          grad_mpi.data_ptr<float>(), pay_ptr, B, S, H, W,
                          bg_inf ? 1 : 0, (int)mode, stream());
   return grad_mpi;
+}
+
+// --------------------------------------------------------------------------
+// nearest x2 upsample — flat logical (N,H,W,C); bf16 needs C%8==0,
+// f32 C%4==0 (see resample_kernels.hip).
+
+at::Tensor upsample2x_fwd(at::Tensor in, int64_t N, int64_t H, int64_t W,
+                          int64_t C) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous() &&
+              in.numel() == N * H * W * C);
+  const bool bf16 = in.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf16 ? (C % 8 == 0) : (C % 4 == 0));
+  auto out = at::empty({N * H * W * 4 * C}, in.options());
+  mine_upsample2x_fwd(in.data_ptr(), out.data_ptr(), N, H, W, C,
+                      bf16 ? 1 : 0, stream());
+  return out;
+}
+
+at::Tensor upsample2x_bwd(at::Tensor gout, int64_t N, int64_t H, int64_t W,
+                          int64_t C) {
+  TORCH_CHECK(gout.is_cuda() && gout.is_contiguous() &&
+              gout.numel() == N * H * W * 4 * C);
+  const bool bf16 = gout.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf16 ? (C % 8 == 0) : (C % 4 == 0));
+  auto gin = at::empty({N * H * W * C}, gout.options());
+  mine_upsample2x_bwd(gout.data_ptr(), gin.data_ptr(), N, H, W, C,
+                      bf16 ? 1 : 0, stream());
+  return gin;
 }
 
 // --------------------------------------------------------------------------
@@ -447,6 +479,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("tgt_composite_bwd", &tgt_composite_bwd);
   mod.def("ssim_fwd", &ssim_fwd);
   mod.def("ssim_bwd", &ssim_bwd);
+  mod.def("upsample2x_fwd", &upsample2x_fwd,
+          "nearest x2 upsample fwd, flat NHWC");
+  mod.def("upsample2x_bwd", &upsample2x_bwd,
+          "nearest x2 upsample bwd (4-child sum), flat NHWC");
   mod.def("reflect_pad_fwd", &reflect_pad_fwd,
           "gather reflection pad over logical (N,H,W,C)");
   mod.def("reflect_pad_bwd", &reflect_pad_bwd,

@@ -550,3 +550,28 @@ def test_tgt_backward_gather_matches_scatter(bg_inf):
                                       bg_inf, g_rgb, g_depth, 1)
     torch.testing.assert_close(gm_gather, gm_scatter, rtol=1e-4, atol=1e-4)
     assert not torch.equal(gm_gather, torch.zeros_like(gm_gather))
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_upsample2x_matches_torch(dtype):
+    import torch.nn.functional as F
+    from mine_amd.ops.upsample import upsample_nearest2x
+
+    g = torch.Generator().manual_seed(5)
+    x0 = torch.randn(3, 16, 9, 26, generator=g)
+    gy0 = torch.randn(3, 16, 18, 52, generator=g)
+
+    x = x0.to("cuda:0", dtype).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    y = upsample_nearest2x(x)
+    assert y.shape == (3, 16, 18, 52) and y.dtype == dtype
+    (y.float() * gy0.cuda()).sum().backward()
+
+    xr = x0.to(dtype).float().requires_grad_(True)
+    yr = F.interpolate(xr, scale_factor=2, mode="nearest")
+    (yr * gy0.to(dtype).float()).sum().backward()
+
+    torch.testing.assert_close(y.float().cpu(), yr.detach(),
+                               rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(x.grad.float().cpu(), xr.grad,
+                               rtol=1e-2, atol=1e-2)

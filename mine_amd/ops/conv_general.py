@@ -1,0 +1,176 @@
+"""General MFMA implicit-GEMM convolution (encoder / neck / base convs).
+
+Carries the ResNet-50 stack (ref network/monodepth2/resnet_encoder.py:
+100-108: 7x7 s2 stem, 1x1 / 3x3 bottleneck convs, 1x1 s2 downsamples),
+the decoder's receptive-field neck (ref depth_decoder.py:56-61) and the
+SplitConvBlock base convs (reflect-padded 3x3, ref monodepth2/
+layers.py:123-138) on the hand-written igemm kernels
+(ops/csrc/igemm_kernels.hip) — forward, data-grad and weight-grad.
+These shapes are tiny (batch 4, L2-resident) and were launch-bound on
+the library path; see the kernel header for the design.
+
+Weights are re-packed to exact MFMA fragment order each call via a
+cached vectorized index gather (microseconds).
+"""
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from mine_amd.ops.backend import get_extension
+
+_LUT: Dict[Tuple, torch.Tensor] = {}
+
+
+def _pack_lut_general(K: int, C: int, R: int, S: int, device) -> torch.Tensor:
+    """Fragment-order gather indices over a flat (K, C, R, S) weight
+    (+ one trailing zero slot). k-ordering: seg = tap*(C/8) + c_oct,
+    element = 8 consecutive channels of one tap (igemm_kernels.hip)."""
+    key = (K, C, R, S, str(device))
+    lut = _LUT.get(key)
+    if lut is not None:
+        return lut
+    Cv = C // 8
+    nseg = R * S * Cv
+    nchunks = (nseg + 3) // 4
+    nK = (K + 15) // 16
+    nc = torch.arange(nK).view(-1, 1, 1, 1)
+    kc = torch.arange(nchunks).view(1, -1, 1, 1)
+    lane = torch.arange(64).view(1, 1, -1, 1)
+    e = torch.arange(8).view(1, 1, 1, -1)
+    seg = kc * 4 + (lane >> 4)
+    kout = nc * 16 + (lane & 15)
+    tap = seg // Cv
+    c = (seg - tap * Cv) * 8 + e
+    idx = (kout * C + c) * (R * S) + tap
+    invalid = (seg >= nseg) | (kout >= K)
+    idx = torch.where(invalid, torch.tensor(K * C * R * S), idx)
+    lut = idx.reshape(-1).to(device)
+    _LUT[key] = lut
+    return lut
+
+
+def pack_weights_general(w: torch.Tensor) -> torch.Tensor:
+    """(K, C, R, S) -> fragment-ordered bf16 buffer."""
+    K, C, R, S = w.shape
+    lut = _pack_lut_general(K, C, R, S, w.device)
+    flat = torch.cat((w.contiguous().reshape(-1), w.new_zeros(1)))
+    return flat.to(torch.bfloat16)[lut].contiguous()
+
+
+def _pad8(w: torch.Tensor) -> torch.Tensor:
+    """Zero-pad in-channels to a multiple of 8 (the 7x7 stem's C=3)."""
+    C = w.shape[1]
+    Cp = (C + 7) & ~7
+    if Cp == C:
+        return w
+    return torch.cat((w, w.new_zeros(w.shape[0], Cp - C, *w.shape[2:])), 1)
+
+
+class _ConvIgemmFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pad, reflect):
+        ext = get_extension(required=True)
+        B, C, Hs, Ws = x.shape
+        K, _, R, S = w.shape
+        if C % 8:
+            x = torch.cat((x, x.new_zeros(B, 8 - C % 8, Hs, Ws)),
+                          1).contiguous(memory_format=torch.channels_last)
+            wq = _pad8(w)
+        else:
+            wq = w
+        Cp = wq.shape[1]
+        if reflect:
+            P, Q = Hs, Ws  # reflect pad keeps size (stride 1, pad (R-1)/2)
+        else:
+            P = (Hs + 2 * pad - R) // stride + 1
+            Q = (Ws + 2 * pad - S) // stride + 1
+        wp = pack_weights_general(wq)
+        M = B * P * Q
+        out = ext.conv_igemm_fwd(
+            x.permute(0, 2, 3, 1).reshape(-1), wp,
+            bias.float() if bias is not None else
+            torch.empty(0, device=x.device, dtype=torch.float32),
+            M, P, Q, K, Hs, Ws, Cp, R, S,
+            stride, 1, -pad, 1, 1 if reflect else 0)
+        ctx.save_for_backward(x, w)
+        ctx.geom = (B, C, Cp, Hs, Ws, K, R, S, P, Q, stride, pad, reflect)
+        ctx.has_bias = bias is not None
+        return out.view(B, P, Q, K).permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, gy):
+        ext = get_extension(required=True)
+        x, w = ctx.saved_tensors  # x already channel-padded
+        B, C, Cp, Hs, Ws, K, R, S, P, Q, stride, pad, reflect = ctx.geom
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        gy_flat = gy.permute(0, 2, 3, 1).reshape(-1)
+        x_flat = x.permute(0, 2, 3, 1).reshape(-1)
+
+        gx = gw = gb = None
+        if ctx.needs_input_grad[1]:
+            # wrw maps OUT pixels through the forward tap coordinates
+            dw = ext.conv_igemm_wrw(
+                x_flat, gy_flat, B * P * Q, P, Q, K, Hs, Ws, Cp, R, S,
+                stride, 1, -pad, 1, 1 if reflect else 0)
+            gw = dw.view(K, Cp, R, S)[:, :C].to(w.dtype)
+        if ctx.has_bias:
+            gb = gy.float().sum((0, 2, 3)).to(w.dtype)
+        if ctx.needs_input_grad[0]:
+            wq = _pad8(w)
+            w_t = wq.permute(1, 0, 2, 3).flip(2, 3)  # (Cp, K, R, S)
+            wtp = pack_weights_general(w_t)
+            empty = torch.empty(0, device=x.device, dtype=torch.float32)
+            if reflect:
+                # grad to the (virtually) padded input, then reflect-fold
+                Hp, Wp_ = Hs + 2 * pad, Ws + 2 * pad
+                gxp = ext.conv_igemm_fwd(
+                    gy_flat, wtp, empty, B * Hp * Wp_, Hp, Wp_, Cp,
+                    P, Q, K, R, S, 1, -1, 0, 1, 0)
+                gx = ext.reflect_pad_bwd(gxp, B, Hs, Ws, Cp, pad)
+                gx = gx.view(B, Hs, Ws, Cp).permute(0, 3, 1, 2)
+            else:
+                gxf = ext.conv_igemm_fwd(
+                    gy_flat, wtp, empty, B * Hs * Ws, Hs, Ws, Cp,
+                    P, Q, K, R, S, 1, -1, pad, stride, 0)
+                gx = gxf.view(B, Hs, Ws, Cp).permute(0, 3, 1, 2)
+            if Cp != C:
+                gx = gx[:, :C]
+        return gx, gw, gb, None, None, None
+
+
+def conv2d_mfma(x: torch.Tensor, w: torch.Tensor,
+                bias: Optional[torch.Tensor] = None, stride: int = 1,
+                padding: int = 0, reflect: bool = False) -> torch.Tensor:
+    """Conv2d on the hand-written igemm kernels; eager fallback when the
+    fast-path conditions don't hold. reflect=True means
+    ReflectionPad((R-1)/2) + unpadded conv (stride must be 1)."""
+    K, C, R, S = w.shape
+    usable = (x.is_cuda and x.dtype == torch.bfloat16
+              and (R, S) in ((1, 1), (3, 3), (7, 7))
+              and K % 8 == 0
+              and x.is_contiguous(memory_format=torch.channels_last)
+              and (not reflect or stride == 1))
+    if usable:
+        return _ConvIgemmFn.apply(x, w, bias,
+                                  stride, padding if not reflect
+                                  else (R - 1) // 2, reflect)
+    if reflect:
+        x = F.pad(x, ((S - 1) // 2,) * 2 + ((R - 1) // 2,) * 2,
+                  mode="reflect")
+        padding = 0
+    return F.conv2d(x, w.to(x.dtype),
+                    bias.to(x.dtype) if bias is not None else None,
+                    stride=stride, padding=padding)
+
+
+class Conv2dMFMA(nn.Conv2d):
+    """nn.Conv2d whose forward runs on the igemm kernels (same
+    parameter/state-dict surface; checkpoint contract unchanged)."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return conv2d_mfma(x, self.weight, self.bias,
+                           stride=self.stride[0], padding=self.padding[0])

@@ -25,6 +25,12 @@ void mine_tgt_composite_bwd(const float*, const float*, const float*,
                             const float*, const float*, const float*,
                             const float*, const float*, float*, float*,
                             int, int, int, int, int, int, hipStream_t);
+void mine_conv_igemm_fwd(const void*, const void*, const float*, void*,
+                         int64_t, int, int, int, int, int, int, int, int,
+                         int, int, int, int, int, hipStream_t);
+void mine_conv_igemm_wrw(const void*, const void*, float*, int64_t, int, int,
+                         int, int, int, int, int, int, int, int, int, int,
+                         int, hipStream_t);
 void mine_upsample2x_fwd(const void*, void*, int64_t, int64_t, int64_t,
                          int64_t, int, hipStream_t);
 void mine_upsample2x_bwd(const void*, void*, int64_t, int64_t, int64_t,
@@ -178,6 +184,45 @@ at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor hfwd,
                          grad_mpi.data_ptr<float>(), pay_ptr, B, S, H, W,
                          bg_inf ? 1 : 0, (int)mode, stream());
   return grad_mpi;
+}
+
+// --------------------------------------------------------------------------
+// general igemm conv (encoder/neck/base shapes; igemm_kernels.hip)
+
+at::Tensor conv_igemm_fwd(at::Tensor x_flat, at::Tensor wp, at::Tensor bias,
+                          int64_t M, int64_t P, int64_t Q, int64_t K,
+                          int64_t Hs, int64_t Ws, int64_t C, int64_t R,
+                          int64_t S, int64_t SA, int64_t SB, int64_t SD,
+                          int64_t SE, int64_t pad_mode) {
+  TORCH_CHECK(x_flat.is_cuda() && x_flat.is_contiguous() &&
+              wp.is_contiguous());
+  TORCH_CHECK(x_flat.scalar_type() == at::kBFloat16 &&
+              wp.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(C % 8 == 0);
+  auto out = at::empty({M * K}, x_flat.options());
+  mine_conv_igemm_fwd(x_flat.data_ptr(), wp.data_ptr(),
+                      bias.numel() ? bias.data_ptr<float>() : nullptr,
+                      out.data_ptr(), M, (int)P, (int)Q, (int)K, (int)Hs,
+                      (int)Ws, (int)C, (int)R, (int)S, (int)SA, (int)SB,
+                      (int)SD, (int)SE, (int)pad_mode, stream());
+  return out;
+}
+
+at::Tensor conv_igemm_wrw(at::Tensor x_flat, at::Tensor gy_flat, int64_t M,
+                          int64_t P, int64_t Q, int64_t K, int64_t Hs,
+                          int64_t Ws, int64_t C, int64_t R, int64_t S,
+                          int64_t SA, int64_t SB, int64_t SD, int64_t SE,
+                          int64_t pad_mode) {
+  TORCH_CHECK(x_flat.is_cuda() && x_flat.is_contiguous() &&
+              gy_flat.is_contiguous());
+  TORCH_CHECK(x_flat.scalar_type() == at::kBFloat16 &&
+              gy_flat.scalar_type() == at::kBFloat16);
+  auto dw = at::zeros({K, C, R, S}, x_flat.options().dtype(at::kFloat));
+  mine_conv_igemm_wrw(x_flat.data_ptr(), gy_flat.data_ptr(),
+                      dw.data_ptr<float>(), M, (int)P, (int)Q, (int)K,
+                      (int)Hs, (int)Ws, (int)C, (int)R, (int)S, (int)SA,
+                      (int)SB, (int)SD, (int)SE, (int)pad_mode, stream());
+  return dw;
 }
 
 // --------------------------------------------------------------------------
@@ -479,6 +524,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("tgt_composite_bwd", &tgt_composite_bwd);
   mod.def("ssim_fwd", &ssim_fwd);
   mod.def("ssim_bwd", &ssim_bwd);
+  mod.def("conv_igemm_fwd", &conv_igemm_fwd,
+          "general igemm conv fwd / data-grad (coordinate-remapped)");
+  mod.def("conv_igemm_wrw", &conv_igemm_wrw,
+          "general igemm conv weight-grad");
   mod.def("upsample2x_fwd", &upsample2x_fwd,
           "nearest x2 upsample fwd, flat NHWC");
   mod.def("upsample2x_bwd", &upsample2x_bwd,

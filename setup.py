@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "conv_kernels.hip"),
         os.path.join(CSRC, "wrw_kernels.hip"),
         os.path.join(CSRC, "resample_kernels.hip"),
+        os.path.join(CSRC, "igemm_kernels.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

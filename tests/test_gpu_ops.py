@@ -575,3 +575,51 @@ def test_upsample2x_matches_torch(dtype):
                                rtol=1e-2, atol=1e-2)
     torch.testing.assert_close(x.grad.float().cpu(), xr.grad,
                                rtol=1e-2, atol=1e-2)
+
+
+@pytest.mark.parametrize("shape", [
+    # (B, C, H, W, K, R, stride, pad, reflect)
+    (4, 64, 16, 24, 256, 1, 1, 0, False),      # bottleneck 1x1
+    (4, 256, 16, 24, 512, 1, 2, 0, False),     # downsample 1x1 s2
+    (4, 64, 16, 24, 64, 3, 1, 1, False),       # bottleneck 3x3
+    (4, 128, 16, 24, 128, 3, 2, 1, False),     # bottleneck 3x3 s2
+    (2, 3, 32, 48, 64, 7, 2, 3, False),        # stem 7x7 s2 C=3 (pad8)
+    (4, 96, 10, 14, 40, 3, 1, 1, True),        # reflect base conv
+    (4, 24, 9, 13, 16, 1, 1, 0, False),        # odd sizes
+])
+def test_conv_igemm_matches_torch(shape):
+    import torch.nn.functional as F
+    from mine_amd.ops.conv_general import conv2d_mfma
+
+    B, C, H, W, K, R, stride, pad, reflect = shape
+    g = torch.Generator().manual_seed(13)
+    x0 = torch.randn(B, C, H, W, generator=g)
+    w0 = torch.randn(K, C, R, R, generator=g) * (0.5 / R)
+    b0 = torch.randn(K, generator=g) * 0.1
+
+    x = x0.to("cuda:0", torch.bfloat16).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    w = w0.cuda().requires_grad_(True)
+    b = b0.cuda().requires_grad_(True)
+    y = conv2d_mfma(x, w, b, stride=stride, padding=pad, reflect=reflect)
+    assert y.dtype == torch.bfloat16
+    gy = torch.randn(y.shape, generator=g)
+    (y.float() * gy.cuda()).sum().backward()
+
+    xq = x0.to(torch.bfloat16).float().requires_grad_(True)
+    wq = w0.to(torch.bfloat16).float().requires_grad_(True)
+    bq = b0.clone().requires_grad_(True)
+    if reflect:
+        yr = F.conv2d(F.pad(xq, (pad,) * 4 if pad else
+                      ((R - 1) // 2,) * 4, mode="reflect"), wq, bq)
+    else:
+        yr = F.conv2d(xq, wq, bq, stride=stride, padding=pad)
+    (yr * gy.to(torch.bfloat16).float()).sum().backward()
+
+    torch.testing.assert_close(y.float().cpu(), yr.detach(),
+                               rtol=5e-2, atol=5e-2)
+    if x.grad is not None:
+        torch.testing.assert_close(x.grad.float().cpu(), xq.grad,
+                                   rtol=5e-2, atol=1e-1)
+    torch.testing.assert_close(w.grad.cpu(), wq.grad, rtol=5e-2, atol=5e-1)
+    torch.testing.assert_close(b.grad.cpu(), bq.grad, rtol=5e-2, atol=5e-1)

@@ -93,8 +93,11 @@ class SplitConvBlock(nn.Module):
         main = torch.cuda.current_stream() if use_side else None
 
         def base_part():
-            # base part at batch B (bias lives here; it is S-invariant)
-            y_base = F.conv2d(self.pad(base), w[:, d:d + b], self.conv.bias)
+            # base part at batch B (bias lives here; it is S-invariant);
+            # reflect pad folded into the igemm kernel's coordinate map
+            from mine_amd.ops.conv_general import conv2d_mfma
+            y_base = conv2d_mfma(base, w[:, d:d + b].contiguous(),
+                                 self.conv.bias, reflect=True)
             K = y_base.shape[1]
             # PE part: conv of a spatially-constant field == channel bias
             w_pe = w[:, d + b:].sum((2, 3))  # (K, E)
@@ -127,8 +130,9 @@ class SplitConvBlock(nn.Module):
 
 def _neck_conv(in_ch: int, out_ch: int, k: int) -> nn.Sequential:
     from mine_amd.ops.bn import FusedBNAct
+    from mine_amd.ops.conv_general import Conv2dMFMA
     return nn.Sequential(
-        nn.Conv2d(in_ch, out_ch, k, stride=1, padding=(k - 1) // 2, bias=False),
+        Conv2dMFMA(in_ch, out_ch, k, stride=1, padding=(k - 1) // 2, bias=False),
         FusedBNAct(out_ch, act="lrelu"),
     )
 

@@ -25,6 +25,7 @@ import torch
 import torch.nn as nn
 
 from mine_amd.ops.bn import FusedBNAct
+from mine_amd.ops.conv_general import Conv2dMFMA
 
 _IMAGENET_MEAN = (0.485, 0.456, 0.406)
 _IMAGENET_STD = (0.229, 0.224, 0.225)
@@ -37,11 +38,11 @@ class Bottleneck(nn.Module):
                  downsample: nn.Module = None):
         super().__init__()
         out_ch = planes * self.expansion
-        self.conv1 = nn.Conv2d(in_ch, planes, 1, bias=False)
+        self.conv1 = Conv2dMFMA(in_ch, planes, 1, bias=False)
         self.bn1 = FusedBNAct(planes, act="relu")
-        self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.conv2 = Conv2dMFMA(planes, planes, 3, stride=stride, padding=1, bias=False)
         self.bn2 = FusedBNAct(planes, act="relu")
-        self.conv3 = nn.Conv2d(planes, out_ch, 1, bias=False)
+        self.conv3 = Conv2dMFMA(planes, out_ch, 1, bias=False)
         # residual join fused into the bn3 epilogue: relu(bn3(conv3) + id)
         self.bn3 = FusedBNAct(out_ch, act="add_relu")
         self.downsample = downsample
@@ -64,7 +65,7 @@ class ResNetEncoder(nn.Module):
         blocks = (3, 4, 6, 3)
         self.num_ch_enc = [64, 256, 512, 1024, 2048]
 
-        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.conv1 = Conv2dMFMA(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = FusedBNAct(64, act="relu")
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
@@ -96,7 +97,7 @@ class ResNetEncoder(nn.Module):
         out_ch = planes * Bottleneck.expansion
         if stride != 1 or self.in_ch != out_ch:
             downsample = nn.Sequential(
-                nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
+                Conv2dMFMA(self.in_ch, out_ch, 1, stride=stride, bias=False),
                 FusedBNAct(out_ch, act="none"),
             )
         layers = [Bottleneck(self.in_ch, planes, stride, downsample)]

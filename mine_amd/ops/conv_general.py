@@ -121,7 +121,9 @@ class _ConvIgemmFn(torch.autograd.Function):
             gb = gy.float().sum((0, 2, 3)).to(w.dtype)
         if ctx.needs_input_grad[0]:
             wq = _pad8(w)
-            w_t = wq.permute(1, 0, 2, 3).flip(2, 3)  # (Cp, K, R, S)
+            # transposed, NOT flipped: the kernel's data-grad coordinate
+            # map (SB = -1) already walks the taps in reverse
+            w_t = wq.permute(1, 0, 2, 3)  # (Cp, K, R, S)
             wtp = pack_weights_general(w_t)
             empty = torch.empty(0, device=x.device, dtype=torch.float32)
             if reflect:

@@ -265,3 +265,91 @@ def test_warp_backward_gather_decomposition():
                                     gather[b, s, qy, qx] += (1 - abs(du)) * \
                                         (1 - abs(dv)) * payload[b, s, ty, tx]
         torch.testing.assert_close(gather, scatter, rtol=1e-4, atol=1e-5)
+
+
+def test_igemm_kernel_coord_and_pack_math():
+    """Simulates conv_igemm_fwd_kernel (ops/csrc/igemm_kernels.hip):
+    fragment seg->(tap, c) decomposition over the general pack LUT and
+    the (SA, SB, SD, SE) source-coordinate map, for forward (stride
+    1/2, zero/reflect pad) AND the data-grad remap, against F.conv2d."""
+    from mine_amd.ops.conv_general import _pack_lut_general
+
+    def reflect(v, n):
+        return _reflect1(v, n)
+
+    def src(p, r, SA, SB, SD, SE, n, pad_mode):
+        num = p * SA + r * SB + SD
+        if SE > 1:
+            if num % SE != 0:
+                return None
+            num //= SE
+        if pad_mode == 1:
+            return reflect(num, n)
+        return num if 0 <= num < n else None
+
+    def run_kernel(x, wfull, P, Q, K, SA, SB, SD, SE, pad_mode, R, S):
+        B, C, Hs, Ws = x.shape
+        Cv = C // 8
+        nseg = R * S * Cv
+        nchunks = (nseg + 3) // 4
+        nK = (K + 15) // 16
+        lut = _pack_lut_general(K, C, R, S, torch.device("cpu"))
+        flat = torch.cat((wfull.reshape(-1), torch.zeros(1)))
+        wp = flat[lut].view(nK, nchunks, 64, 8)
+        xn = x.permute(0, 2, 3, 1)
+        out = torch.zeros(B * P * Q, K)
+        for m in range(B * P * Q):
+            n, rem = divmod(m, P * Q)
+            p, q = divmod(rem, Q)
+            for kc in range(nchunks):
+                for sub in range(4):
+                    seg = kc * 4 + sub
+                    if seg >= nseg:
+                        continue
+                    tap = seg // Cv
+                    coct = seg - tap * Cv
+                    r, s = divmod(tap, S)
+                    ys = src(p, r, SA, SB, SD, SE, Hs, pad_mode)
+                    xs = src(q, s, SA, SB, SD, SE, Ws, pad_mode)
+                    if ys is None or xs is None:
+                        continue
+                    a = xn[n, ys, xs, coct * 8:coct * 8 + 8]
+                    for nc in range(nK):
+                        for j in range(16):
+                            kout = nc * 16 + j
+                            if kout >= K:
+                                continue
+                            b = wp[nc, kc, sub * 16 + j]
+                            out[m, kout] += (a * b).sum()
+        return out.view(B, P, Q, K).permute(0, 3, 1, 2)
+
+    torch.manual_seed(9)
+    B, C, H, W = 1, 8, 5, 6
+    # fwd: 3x3 stride 1 pad 1 (zero), K=20 (ragged)
+    x = torch.randn(B, C, H, W)
+    w = torch.randn(20, C, 3, 3)
+    got = run_kernel(x, w, H, W, 20, 1, 1, -1, 1, 0, 3, 3)
+    torch.testing.assert_close(got, F.conv2d(x, w, padding=1),
+                               rtol=1e-4, atol=1e-4)
+    # fwd: 3x3 stride 2 pad 1
+    got = run_kernel(x, w, (H - 1) // 2 + 1, (W - 1) // 2 + 1, 20,
+                     2, 1, -1, 1, 0, 3, 3)
+    torch.testing.assert_close(got, F.conv2d(x, w, stride=2, padding=1),
+                               rtol=1e-4, atol=1e-4)
+    # fwd: reflect pad 1
+    got = run_kernel(x, w, H, W, 20, 1, 1, -1, 1, 1, 3, 3)
+    torch.testing.assert_close(
+        got, F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"), w),
+        rtol=1e-4, atol=1e-4)
+    # data-grad: stride 2 pad 1 (transposed weights, NOT flipped)
+    P, Q = (H - 1) // 2 + 1, (W - 1) // 2 + 1
+    gy = torch.randn(B, 20, P, Q)
+    w_t = w.permute(1, 0, 2, 3).contiguous()
+    # pad K-channel dim (20) for the LUT's 8-divisibility? 20 % 8 != 0:
+    # pad gy and w_t to 24 channels like the wrapper does
+    gy24 = torch.cat((gy, torch.zeros(B, 4, P, Q)), 1)
+    w_t24 = torch.cat((w_t, torch.zeros(C, 4, 3, 3)), 1)
+    got = run_kernel(gy24, w_t24, H, W, C, 1, -1, 1, 2, 0, 3, 3)
+    xr = x.clone().requires_grad_(True)
+    (F.conv2d(xr, w, stride=2, padding=1) * gy).sum().backward()
+    torch.testing.assert_close(got, xr.grad, rtol=1e-4, atol=1e-4)

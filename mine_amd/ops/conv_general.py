@@ -9,8 +9,9 @@ layers.py:123-138) on the hand-written igemm kernels
 These shapes are tiny (batch 4, L2-resident) and were launch-bound on
 the library path; see the kernel header for the design.
 
-Weights are re-packed to exact MFMA fragment order each call via a
-cached vectorized index gather (microseconds).
+Weight packing is ONE kernel launch: a cached device LUT encodes the
+fragment order AND any transpose (data-grad) or channel padding (the
+stem's C=3), so no permute/pad/cast chain ever materializes.
 """
 from __future__ import annotations
 
@@ -25,49 +26,63 @@ from mine_amd.ops.backend import get_extension
 _LUT: Dict[Tuple, torch.Tensor] = {}
 
 
-def _pack_lut_general(K: int, C: int, R: int, S: int, device) -> torch.Tensor:
-    """Fragment-order gather indices over a flat (K, C, R, S) weight
-    (+ one trailing zero slot). k-ordering: seg = tap*(C/8) + c_oct,
-    element = 8 consecutive channels of one tap (igemm_kernels.hip)."""
-    key = (K, C, R, S, str(device))
+def _frag_lut(K_log: int, C_log: int, R: int, S: int, C_phys: int,
+              K_avail: int, C_avail: int, trans: bool,
+              device) -> torch.Tensor:
+    """Fragment-order gather LUT (int32; -1 -> packed zero).
+
+    Logical weight L[a, b, tap] with a in [0, K_log) rows and b in
+    [0, C_log) contraction channels (C_log % 8 == 0). Physical index
+    into the flat (K_phys, C_phys, R, S) tensor:
+      trans=False: ((a * C_phys + b) * R*S + tap), valid a<K_avail, b<C_avail
+      trans=True:  ((b * C_phys + a) * R*S + tap), valid a<C_avail, b<K_avail
+    (the data-grad uses the TRANSPOSED, unflipped weight — the kernel's
+    coordinate map walks the taps in reverse).
+    """
+    key = (K_log, C_log, R, S, C_phys, K_avail, C_avail, trans, str(device))
     lut = _LUT.get(key)
     if lut is not None:
         return lut
-    Cv = C // 8
+    Cv = C_log // 8
     nseg = R * S * Cv
     nchunks = (nseg + 3) // 4
-    nK = (K + 15) // 16
+    nK = (K_log + 15) // 16
     nc = torch.arange(nK).view(-1, 1, 1, 1)
     kc = torch.arange(nchunks).view(1, -1, 1, 1)
     lane = torch.arange(64).view(1, 1, -1, 1)
     e = torch.arange(8).view(1, 1, 1, -1)
     seg = kc * 4 + (lane >> 4)
-    kout = nc * 16 + (lane & 15)
+    a = (nc * 16 + (lane & 15)).expand(nK, nchunks, 64, 8)
     tap = seg // Cv
-    c = (seg - tap * Cv) * 8 + e
-    idx = (kout * C + c) * (R * S) + tap
-    invalid = (seg >= nseg) | (kout >= K)
-    idx = torch.where(invalid, torch.tensor(K * C * R * S), idx)
-    lut = idx.reshape(-1).to(device)
+    b = ((seg - tap * Cv) * 8 + e).expand(nK, nchunks, 64, 8)
+    tap = tap.expand(nK, nchunks, 64, 8)
+    if trans:
+        idx = (b * C_phys + a) * (R * S) + tap
+        invalid = (seg >= nseg).expand_as(idx) | (a >= C_avail) | \
+                  (b >= K_avail)
+    else:
+        idx = (a * C_phys + b) * (R * S) + tap
+        invalid = (seg >= nseg).expand_as(idx) | (a >= K_avail) | \
+                  (b >= C_avail)
+    idx = torch.where(invalid, torch.tensor(-1), idx)
+    lut = idx.reshape(-1).to(device=device, dtype=torch.int32)
     _LUT[key] = lut
     return lut
 
 
-def pack_weights_general(w: torch.Tensor) -> torch.Tensor:
-    """(K, C, R, S) -> fragment-ordered bf16 buffer."""
+def pack_weights_general(w: torch.Tensor, trans: bool = False) -> torch.Tensor:
+    """(K, C, R, S) contiguous -> fragment-ordered bf16 buffer (one
+    launch). trans packs the transposed (C-major) logical layout for the
+    data-grad."""
+    ext = get_extension(required=True)
     K, C, R, S = w.shape
-    lut = _pack_lut_general(K, C, R, S, w.device)
-    flat = torch.cat((w.contiguous().reshape(-1), w.new_zeros(1)))
-    return flat.to(torch.bfloat16)[lut].contiguous()
-
-
-def _pad8(w: torch.Tensor) -> torch.Tensor:
-    """Zero-pad in-channels to a multiple of 8 (the 7x7 stem's C=3)."""
-    C = w.shape[1]
     Cp = (C + 7) & ~7
-    if Cp == C:
-        return w
-    return torch.cat((w, w.new_zeros(w.shape[0], Cp - C, *w.shape[2:])), 1)
+    Kp = (K + 7) & ~7
+    if trans:
+        lut = _frag_lut(Cp, Kp, R, S, C, K, C, True, w.device)
+    else:
+        lut = _frag_lut(K, Cp, R, S, C, K, C, False, w.device)
+    return ext.pack_gather(w.reshape(-1), lut)
 
 
 class _ConvIgemmFn(torch.autograd.Function):
@@ -79,16 +94,13 @@ class _ConvIgemmFn(torch.autograd.Function):
         if C % 8:
             x = torch.cat((x, x.new_zeros(B, 8 - C % 8, Hs, Ws)),
                           1).contiguous(memory_format=torch.channels_last)
-            wq = _pad8(w)
-        else:
-            wq = w
-        Cp = wq.shape[1]
+        Cp = x.shape[1]
         if reflect:
             P, Q = Hs, Ws  # reflect pad keeps size (stride 1, pad (R-1)/2)
         else:
             P = (Hs + 2 * pad - R) // stride + 1
             Q = (Ws + 2 * pad - S) // stride + 1
-        wp = pack_weights_general(wq)
+        wp = pack_weights_general(w)
         M = B * P * Q
         out = ext.conv_igemm_fwd(
             x.permute(0, 2, 3, 1).reshape(-1), wp,
@@ -116,28 +128,27 @@ class _ConvIgemmFn(torch.autograd.Function):
             dw = ext.conv_igemm_wrw(
                 x_flat, gy_flat, B * P * Q, P, Q, K, Hs, Ws, Cp, R, S,
                 stride, 1, -pad, 1, 1 if reflect else 0)
-            gw = dw.view(K, Cp, R, S)[:, :C].to(w.dtype)
+            gw = dw.view(K, Cp, R, S)[:, :C] if Cp != C \
+                else dw.view(K, C, R, S)
+            gw = gw.to(w.dtype)
         if ctx.has_bias:
             gb = gy.float().sum((0, 2, 3)).to(w.dtype)
         if ctx.needs_input_grad[0]:
-            wq = _pad8(w)
-            # transposed, NOT flipped: the kernel's data-grad coordinate
-            # map (SB = -1) already walks the taps in reverse
-            w_t = wq.permute(1, 0, 2, 3)  # (Cp, K, R, S)
-            wtp = pack_weights_general(w_t)
+            Kp = (K + 7) & ~7
+            wtp = pack_weights_general(w, trans=True)
             empty = torch.empty(0, device=x.device, dtype=torch.float32)
             if reflect:
                 # grad to the (virtually) padded input, then reflect-fold
                 Hp, Wp_ = Hs + 2 * pad, Ws + 2 * pad
                 gxp = ext.conv_igemm_fwd(
                     gy_flat, wtp, empty, B * Hp * Wp_, Hp, Wp_, Cp,
-                    P, Q, K, R, S, 1, -1, 0, 1, 0)
+                    P, Q, Kp, R, S, 1, -1, 0, 1, 0)
                 gx = ext.reflect_pad_bwd(gxp, B, Hs, Ws, Cp, pad)
                 gx = gx.view(B, Hs, Ws, Cp).permute(0, 3, 1, 2)
             else:
                 gxf = ext.conv_igemm_fwd(
                     gy_flat, wtp, empty, B * Hs * Ws, Hs, Ws, Cp,
-                    P, Q, K, R, S, 1, -1, pad, stride, 0)
+                    P, Q, Kp, R, S, 1, -1, pad, stride, 0)
                 gx = gxf.view(B, Hs, Ws, Cp).permute(0, 3, 1, 2)
             if Cp != C:
                 gx = gx[:, :C]
@@ -153,7 +164,8 @@ def conv2d_mfma(x: torch.Tensor, w: torch.Tensor,
     K, C, R, S = w.shape
     usable = (x.is_cuda and x.dtype == torch.bfloat16
               and (R, S) in ((1, 1), (3, 3), (7, 7))
-              and K % 8 == 0
+              and K % 8 == 0  # gy rows feed 8-wide loads in the data-grad
+              and w.is_contiguous()
               and x.is_contiguous(memory_format=torch.channels_last)
               and (not reflect or stride == 1))
     if usable:

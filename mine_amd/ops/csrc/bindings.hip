@@ -31,6 +31,8 @@ void mine_conv_igemm_fwd(const void*, const void*, const float*, void*,
 void mine_conv_igemm_wrw(const void*, const void*, float*, int64_t, int, int,
                          int, int, int, int, int, int, int, int, int, int,
                          int, hipStream_t);
+void mine_pack_gather(const void*, const int*, void*, int64_t, int,
+                      hipStream_t);
 void mine_upsample2x_fwd(const void*, void*, int64_t, int64_t, int64_t,
                          int64_t, int, hipStream_t);
 void mine_upsample2x_bwd(const void*, void*, int64_t, int64_t, int64_t,
@@ -188,6 +190,19 @@ at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor hfwd,
 
 // --------------------------------------------------------------------------
 // general igemm conv (encoder/neck/base shapes; igemm_kernels.hip)
+
+// one-launch fragment pack: bf16 gather through a device LUT (neg -> 0)
+at::Tensor pack_gather(at::Tensor w, at::Tensor lut) {
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous() && lut.is_contiguous());
+  TORCH_CHECK(lut.scalar_type() == at::kInt);
+  TORCH_CHECK(w.scalar_type() == at::kFloat ||
+              w.scalar_type() == at::kBFloat16);
+  auto out = at::empty({lut.numel()}, w.options().dtype(at::kBFloat16));
+  mine_pack_gather(w.data_ptr(), lut.data_ptr<int>(), out.data_ptr(),
+                   lut.numel(), w.scalar_type() == at::kFloat ? 1 : 0,
+                   stream());
+  return out;
+}
 
 at::Tensor conv_igemm_fwd(at::Tensor x_flat, at::Tensor wp, at::Tensor bias,
                           int64_t M, int64_t P, int64_t Q, int64_t K,
@@ -524,6 +539,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("tgt_composite_bwd", &tgt_composite_bwd);
   mod.def("ssim_fwd", &ssim_fwd);
   mod.def("ssim_bwd", &ssim_bwd);
+  mod.def("pack_gather", &pack_gather,
+          "one-launch fragment pack through a device LUT");
   mod.def("conv_igemm_fwd", &conv_igemm_fwd,
           "general igemm conv fwd / data-grad (coordinate-remapped)");
   mod.def("conv_igemm_wrw", &conv_igemm_wrw,

@@ -276,7 +276,38 @@ conv_igemm_wrw_kernel(const __hip_bfloat16* __restrict__ x,   // (N,Hs,Ws,C)
   }
 }
 
+// ---------------------------------------------------------------------------
+// fragment pack: one fused gather+cast (index < 0 -> zero). The LUT
+// already encodes transposes / channel offsets / padding, so a pack is
+// ONE kernel launch regardless of the weight's logical layout.
+// ---------------------------------------------------------------------------
+
+template <typename TIN>
+__global__ void __launch_bounds__(kBlock)
+pack_gather_kernel(const TIN* __restrict__ w, const int* __restrict__ lut,
+                   __hip_bfloat16* __restrict__ out, int64_t n) {
+  const int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= n) return;
+  const int idx = lut[i];
+  out[i] = (idx >= 0) ? (__hip_bfloat16)(float)w[idx] : (__hip_bfloat16)0.0f;
+}
+
 }  // namespace
+
+extern "C" void mine_pack_gather(const void* w, const int* lut, void* out,
+                                 int64_t n, int is_fp32,
+                                 hipStream_t stream) {
+  const dim3 grid((unsigned)((n + kBlock - 1) / kBlock));
+  if (is_fp32)
+    hipLaunchKernelGGL(pack_gather_kernel<float>, grid, dim3(kBlock), 0,
+                       stream, reinterpret_cast<const float*>(w), lut,
+                       reinterpret_cast<__hip_bfloat16*>(out), n);
+  else
+    hipLaunchKernelGGL(pack_gather_kernel<__hip_bfloat16>, grid,
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(w), lut,
+                       reinterpret_cast<__hip_bfloat16*>(out), n);
+}
 
 // ---------------------------------------------------------------------------
 // launchers

@@ -272,7 +272,7 @@ def test_igemm_kernel_coord_and_pack_math():
     fragment seg->(tap, c) decomposition over the general pack LUT and
     the (SA, SB, SD, SE) source-coordinate map, for forward (stride
     1/2, zero/reflect pad) AND the data-grad remap, against F.conv2d."""
-    from mine_amd.ops.conv_general import _pack_lut_general
+    from mine_amd.ops.conv_general import _frag_lut
 
     def reflect(v, n):
         return _reflect1(v, n)
@@ -293,9 +293,12 @@ def test_igemm_kernel_coord_and_pack_math():
         nseg = R * S * Cv
         nchunks = (nseg + 3) // 4
         nK = (K + 15) // 16
-        lut = _pack_lut_general(K, C, R, S, torch.device("cpu"))
-        flat = torch.cat((wfull.reshape(-1), torch.zeros(1)))
-        wp = flat[lut].view(nK, nchunks, 64, 8)
+        Cp = (C + 7) & ~7
+        lut = _frag_lut(K, Cp, R, S, C, K, C, False,
+                        torch.device("cpu")).long()
+        flat = wfull.reshape(-1)
+        wp = torch.where(lut >= 0, flat[lut.clamp(min=0)],
+                         torch.zeros(())).view(nK, nchunks, 64, 8)
         xn = x.permute(0, 2, 3, 1)
         out = torch.zeros(B * P * Q, K)
         for m in range(B * P * Q):

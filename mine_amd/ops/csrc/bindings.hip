@@ -26,8 +26,8 @@ void mine_tgt_composite_bwd(const float*, const float*, const float*,
                             const float*, const float*, float*, float*,
                             int, int, int, int, int, int, hipStream_t);
 void mine_conv_igemm_fwd(const void*, const void*, const float*, void*,
-                         int64_t, int, int, int, int, int, int, int, int,
-                         int, int, int, int, int, hipStream_t);
+                         float*, int64_t, int, int, int, int, int, int, int,
+                         int, int, int, int, int, int, int, hipStream_t);
 void mine_conv_igemm_wrw(const void*, const void*, float*, int64_t, int, int,
                          int, int, int, int, int, int, int, int, int, int,
                          int, hipStream_t);
@@ -215,11 +215,27 @@ at::Tensor conv_igemm_fwd(at::Tensor x_flat, at::Tensor wp, at::Tensor bias,
               wp.scalar_type() == at::kBFloat16);
   TORCH_CHECK(C % 8 == 0);
   auto out = at::empty({M * K}, x_flat.options());
+  // contraction split-K when the (M, K) grid underfills the 256 CUs
+  const int64_t blocks = ((M + 63) / 64) * ((K + 63) / 64);
+  const int nchunks = (int)((R * S * (C / 8) + 3) / 4);
+  int splitz = 1;
+  at::Tensor ws;
+  float* ws_ptr = nullptr;
+  if (blocks < 192 && nchunks >= 16) {
+    splitz = (int)(256 / blocks + 1);
+    if (splitz > nchunks / 4) splitz = (int)(nchunks / 4);
+    if (splitz > 16) splitz = 16;
+    if (splitz > 1) {
+      ws = at::zeros({M * K}, x_flat.options().dtype(at::kFloat));
+      ws_ptr = ws.data_ptr<float>();
+    }
+  }
   mine_conv_igemm_fwd(x_flat.data_ptr(), wp.data_ptr(),
                       bias.numel() ? bias.data_ptr<float>() : nullptr,
-                      out.data_ptr(), M, (int)P, (int)Q, (int)K, (int)Hs,
-                      (int)Ws, (int)C, (int)R, (int)S, (int)SA, (int)SB,
-                      (int)SD, (int)SE, (int)pad_mode, stream());
+                      out.data_ptr(), ws_ptr, M, (int)P, (int)Q, (int)K,
+                      (int)Hs, (int)Ws, (int)C, (int)R, (int)S, (int)SA,
+                      (int)SB, (int)SD, (int)SE, (int)pad_mode, splitz,
+                      stream());
   return out;
 }
 

@@ -72,12 +72,17 @@ __device__ __forceinline__ int src_coord(int p, int r, int SA, int SB,
 // RS = R*S (compile-time: 1, 9, 49) so tap decomposition has constant
 // divisors; Cv = C/8 runtime.
 
-template <int Sdim, int RS, int PAD>
+// SPLIT: gridDim.z slices the contraction (deep-K, tiny-M shapes — the
+// decoder's 2048-channel base convs run at M=384 pixels, 24 blocks
+// without it); slices atomically accumulate into an fp32 workspace
+// that igemm_epilogue_kernel then bias-adds and casts.
+template <int Sdim, int RS, int PAD, bool SPLIT>
 __global__ void __launch_bounds__(kBlock)
 conv_igemm_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,Hs,Ws,C)
                       const __hip_bfloat16* __restrict__ wp, // packed frags
                       const float* __restrict__ bias,        // (K) or null
                       __hip_bfloat16* __restrict__ out,      // (M,K) flat
+                      float* __restrict__ ws_out,            // (M,K) fp32
                       int64_t M, int P, int Q, int K,
                       int Hs, int Ws, int C,
                       int SA, int SB, int SD, int SE) {
@@ -103,11 +108,17 @@ conv_igemm_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,Hs,Ws,C)
   const int nK = (K + 15) / 16;
   const int nk_here = (k0 + 64 <= K) ? 4 : (nK - blockIdx.y * 4);
 
+  int kc_begin = 0, kc_end = nchunks;
+  if (SPLIT) {
+    kc_begin = (int)((int64_t)nchunks * blockIdx.z / gridDim.z);
+    kc_end = (int)((int64_t)nchunks * (blockIdx.z + 1) / gridDim.z);
+  }
+
   f32x4 acc[4];
 #pragma unroll
   for (int a = 0; a < 4; ++a) acc[a] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (int kc = 0; kc < nchunks; ++kc) {
+  for (int kc = kc_begin; kc < kc_end; ++kc) {
     const int seg = kc * 4 + (lane >> 4);
     bf16x8 afrag;
     bool loaded = false;
@@ -154,10 +165,25 @@ conv_igemm_fwd_kernel(const __hip_bfloat16* __restrict__ x,  // (N,Hs,Ws,C)
     for (int rr = 0; rr < 4; ++rr) {
       const int64_t mm = m_out + rr;
       if (mm < M) {
-        out[mm * K + kout] = (__hip_bfloat16)(acc[a][rr] + b);
+        if (SPLIT) {
+          atomicAdd(&ws_out[mm * K + kout], acc[a][rr]);
+        } else {
+          out[mm * K + kout] = (__hip_bfloat16)(acc[a][rr] + b);
+        }
       }
     }
   }
+}
+
+__global__ void __launch_bounds__(kBlock)
+igemm_epilogue_kernel(const float* __restrict__ ws,
+                      const float* __restrict__ bias,
+                      __hip_bfloat16* __restrict__ out, int64_t total,
+                      int K) {
+  const int64_t i = (int64_t)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= total) return;
+  const float b = bias ? bias[(int)(i % K)] : 0.0f;
+  out[i] = (__hip_bfloat16)(ws[i] + b);
 }
 
 // ---------------------------------------------------------------------------
@@ -324,29 +350,40 @@ extern "C" void mine_pack_gather(const void* w, const int* lut, void* out,
     KERNEL_CALL;                                                          \
   }
 
+// splitz > 1 requires ws (pre-zeroed fp32 (M,K)); the epilogue launch
+// bias-adds and casts into out.
 extern "C" void mine_conv_igemm_fwd(
-    const void* x, const void* wp, const float* bias, void* out,
+    const void* x, const void* wp, const float* bias, void* out, float* ws,
     int64_t M, int P, int Q, int K, int Hs, int Ws, int C,
-    int R, int S, int SA, int SB, int SD, int SE, int pad_mode,
+    int R, int S, int SA, int SB, int SD, int SE, int pad_mode, int splitz,
     hipStream_t stream) {
-  const int Cv = C / 8;
-  const int nchunks = (R * S * Cv + 3) / 4;
-  (void)nchunks;
-  const dim3 grid((unsigned)((M + 63) / 64), (unsigned)((K + 63) / 64));
-#define CALL_FWD                                                         \
-  hipLaunchKernelGGL((conv_igemm_fwd_kernel<Sdim, RS, PAD>), grid,       \
-                     dim3(kBlock), 0, stream,                            \
+  const dim3 grid((unsigned)((M + 63) / 64), (unsigned)((K + 63) / 64),
+                  (unsigned)(splitz > 1 ? splitz : 1));
+#define CALL_FWD(SPLIT)                                                  \
+  hipLaunchKernelGGL((conv_igemm_fwd_kernel<Sdim, RS, PAD, SPLIT>),      \
+                     grid, dim3(kBlock), 0, stream,                      \
                      reinterpret_cast<const __hip_bfloat16*>(x),         \
                      reinterpret_cast<const __hip_bfloat16*>(wp), bias,  \
-                     reinterpret_cast<__hip_bfloat16*>(out),             \
+                     reinterpret_cast<__hip_bfloat16*>(out), ws,         \
                      M, P, Q, K, Hs, Ws, C, SA, SB, SD, SE)
+#define CALL_FWD_EITHER                                                  \
+  do { if (splitz > 1) { CALL_FWD(true); } else { CALL_FWD(false); } }   \
+  while (0)
   if (R == 1 && S == 1) {
-    IGEMM_DISPATCH_RS(1, 1, pad_mode, CALL_FWD)
+    IGEMM_DISPATCH_RS(1, 1, pad_mode, CALL_FWD_EITHER)
   } else if (R == 3 && S == 3) {
-    IGEMM_DISPATCH_RS(3, 9, pad_mode, CALL_FWD)
+    IGEMM_DISPATCH_RS(3, 9, pad_mode, CALL_FWD_EITHER)
   } else if (R == 7 && S == 7) {
-    IGEMM_DISPATCH_RS(7, 49, pad_mode, CALL_FWD)
+    IGEMM_DISPATCH_RS(7, 49, pad_mode, CALL_FWD_EITHER)
   }
+  if (splitz > 1) {
+    const int64_t total = M * K;
+    hipLaunchKernelGGL(igemm_epilogue_kernel,
+                       dim3((unsigned)((total + kBlock - 1) / kBlock)),
+                       dim3(kBlock), 0, stream, ws, bias,
+                       reinterpret_cast<__hip_bfloat16*>(out), total, K);
+  }
+#undef CALL_FWD_EITHER
 #undef CALL_FWD
 }
 

@@ -1,76 +1,98 @@
-"""Minimal dispatch set for PMC counter collection: each hand-written
-hot kernel exactly a few times at flagship-like shapes (full-bench PMC
-serializes thousands of dispatches and takes forever)."""
+"""Minimal dispatch set for PMC counter collection — direct extension
+calls, ~40 dispatches total (counter mode costs ~0.1-1 s per dispatch)."""
 import os, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
-import torch.nn.functional as F
 
 dev = "cuda:0"
 torch.manual_seed(0)
+from mine_amd.ops.backend import get_extension
+ext = get_extension(required=True)
 
-# conv3x3 fused reflect (decoder hot conv) fwd + wrw + bwd-data
-from mine_amd.ops.conv import conv3x3_reflect
-x = torch.randn(256, 16, 256, 384, device=dev, dtype=torch.bfloat16
-                ).contiguous(memory_format=torch.channels_last
-                ).requires_grad_(True)
-w = (torch.randn(16, 16, 3, 3, device=dev) * 0.2).requires_grad_(True)
-for _ in range(2):
-    y = conv3x3_reflect(x, w, None)
-    y.float().sum().backward()
-    x.grad = w.grad = None
+def cl(t):
+    return t.contiguous(memory_format=torch.channels_last)
 
-# general igemm (encoder bottleneck 3x3 + deep base conv)
-from mine_amd.ops.conv_general import conv2d_mfma
-xe = torch.randn(4, 256, 16, 24, device=dev, dtype=torch.bfloat16
-                 ).contiguous(memory_format=torch.channels_last
-                 ).requires_grad_(True)
-we = (torch.randn(256, 256, 3, 3, device=dev) * 0.05).requires_grad_(True)
-for _ in range(2):
-    conv2d_mfma(xe, we, None, padding=1).float().sum().backward()
-    xe.grad = we.grad = None
+# --- conv3x3 fused reflect fwd + wrw + bwd-data (decoder hot conv) ---
+from mine_amd.ops.conv import pack_weights, conv3x3_bwd_data
+B, C, H, W, K = 64, 16, 128, 192, 16
+x = cl(torch.randn(B, C, H, W, device=dev, dtype=torch.bfloat16))
+w = torch.randn(K, C, 3, 3, device=dev) * 0.2
+gy = cl(torch.randn(B, K, H, W, device=dev, dtype=torch.bfloat16))
+wp = pack_weights(w.to(torch.bfloat16))
+out = torch.empty(B * H * W * K, device=dev, dtype=torch.bfloat16)
+ext.conv3x3_fwd(x.permute(0, 2, 3, 1).reshape(-1), wp,
+                torch.empty(0, device=dev), out, B, H, W, C, K, 0, H, W, 0)
+ext.conv3x3_wrw(x.permute(0, 2, 3, 1).reshape(-1),
+                gy.permute(0, 2, 3, 1).reshape(-1), B, H, W, C, K)
+conv3x3_bwd_data(gy, w)
 
-# fused BN+ELU fwd+bwd
-from mine_amd.ops.bn import FusedBNAct
-bn = FusedBNAct(16, act="elu").to(dev).train()
-xb = torch.randn(256, 16, 256, 384, device=dev, dtype=torch.bfloat16
-                 ).contiguous(memory_format=torch.channels_last
-                 ).requires_grad_(True)
-for _ in range(2):
-    bn(xb).float().sum().backward()
-    xb.grad = None
+# --- general igemm fwd/wrw (encoder bottleneck) ---
+from mine_amd.ops.conv_general import pack_weights_general
+xe = cl(torch.randn(4, 256, 16, 24, device=dev, dtype=torch.bfloat16))
+we = torch.randn(256, 256, 3, 3, device=dev) * 0.05
+gye = cl(torch.randn(4, 256, 16, 24, device=dev, dtype=torch.bfloat16))
+wep = pack_weights_general(we)
+M = 4 * 16 * 24
+ext.conv_igemm_fwd(xe.permute(0, 2, 3, 1).reshape(-1), wep,
+                   torch.empty(0, device=dev), M, 16, 24, 256, 16, 24, 256,
+                   3, 3, 1, 1, -1, 1, 0)
+ext.conv_igemm_wrw(xe.permute(0, 2, 3, 1).reshape(-1),
+                   gye.permute(0, 2, 3, 1).reshape(-1), M, 16, 24, 256,
+                   16, 24, 256, 3, 3, 1, 1, -1, 1, 0)
 
-# fused renderers fwd+bwd (src + tgt incl. gather)
-from mine_amd.ops.renderer import pack_mpi, render_src_view, render_tgt_view
-B, S, H, W = 4, 64, 256, 384
-rgb = torch.rand(B, S, 3, H, W, device=dev)
-sig = torch.rand(B, S, 1, H, W, device=dev) * 3 + 1e-4
-disp, _ = torch.sort(torch.rand(B, S, device=dev) * 0.9 + 0.05, dim=1,
+# --- fused BN kernels ---
+xb = cl(torch.randn(B, C, H, W, device=dev, dtype=torch.bfloat16))
+xb_flat = xb.permute(0, 2, 3, 1).reshape(-1)
+Mb = B * H * W
+mean, invstd = ext.bn_stats(xb_flat, Mb, C, torch.empty(0, device=dev),
+                            torch.empty(0, device=dev), 1e-5, 0.1)
+gamma = torch.ones(C, device=dev)
+beta = torch.zeros(C, device=dev)
+empty_b = torch.empty(0, device=dev, dtype=torch.bfloat16)
+y = ext.bn_act_fwd(xb_flat, empty_b, mean, invstd, gamma, beta, Mb, C, 3)
+gyb = gy.permute(0, 2, 3, 1).reshape(-1)
+red = ext.bn_act_bwd_reduce(xb_flat, empty_b, gyb, mean, invstd, gamma,
+                            beta, Mb, C, 3)
+ext.bn_act_bwd_dx(xb_flat, empty_b, gyb, mean, invstd, gamma, beta, red,
+                  Mb, C, 3, Mb)
+
+# --- renderers (src + tgt fwd, tgt bwd gather mode) ---
+from mine_amd.ops import torch_ref as tr
+from mine_amd.utils.geometry import inverse_3x3
+Bb, S, Hh, Ww = 2, 32, 128, 192
+mpi = torch.rand(Bb, S, Hh, Ww, 4, device=dev)
+disp, _ = torch.sort(torch.rand(Bb, S, device=dev) * 0.9 + 0.05, dim=1,
                      descending=True)
-f = 0.8 * W
-K = torch.tensor([[f, 0, W / 2], [0, f, H / 2], [0, 0, 1.0]],
-                 device=dev).unsqueeze(0).repeat(B, 1, 1)
-K_inv = torch.inverse(K)
-G = torch.eye(4, device=dev).unsqueeze(0).repeat(B, 1, 1)
+depths = torch.reciprocal(disp).contiguous()
+f = 0.8 * Ww
+Km = torch.tensor([[f, 0, Ww / 2], [0, f, Hh / 2], [0, 0, 1.0]],
+                  device=dev).unsqueeze(0).repeat(Bb, 1, 1)
+K_inv = torch.inverse(Km)
+G = torch.eye(4, device=dev).unsqueeze(0).repeat(Bb, 1, 1)
 G[:, 0, 3] = 0.1
-img = torch.rand(B, 3, H, W, device=dev)
-mpi = pack_mpi(rgb, sig).requires_grad_(True)
-r, d, blend = render_src_view(mpi, disp, K_inv, src_img=img)
-tr_, td, tm = render_tgt_view(blend, disp, G, K_inv, K)
-(r.sum() + tr_.sum() + td.sum()).backward()
+img = torch.rand(Bb, Hh, Ww, 3, device=dev)
+ext.src_composite_fwd(mpi, depths, K_inv.contiguous(), img, False)
+hinv = tr.homography_tgt_to_src(G, depths, K_inv, Km).contiguous()
+m = torch.matmul(G[:, :3, :3], K_inv).contiguous()
+tvec = G[:, :3, 3].contiguous()
+ext.tgt_composite_fwd(mpi, hinv, m, tvec, depths, False)
+hfwd = inverse_3x3(hinv.reshape(-1, 3, 3)).reshape(Bb, S, 3, 3).contiguous()
+g_rgb = torch.rand(Bb, 3, Hh, Ww, device=dev)
+g_depth = torch.rand(Bb, 1, Hh, Ww, device=dev)
+ext.tgt_composite_bwd(mpi, hinv, hfwd, m, tvec, depths, False, g_rgb,
+                      g_depth, 1)
 
-# SSIM fwd+bwd
+# --- SSIM fwd ---
 from mine_amd.ops.ssim import ssim
-a = torch.rand(4, 3, 256, 384, device=dev).requires_grad_(True)
-b = torch.rand(4, 3, 256, 384, device=dev)
-(1 - ssim(a, b)).backward()
+a = torch.rand(4, 3, 128, 192, device=dev)
+b = torch.rand(4, 3, 128, 192, device=dev)
+with torch.no_grad():
+    ssim(a, b)
 
-# upsample
-from mine_amd.ops.upsample import upsample_nearest2x
-xu = torch.randn(256, 16, 128, 192, device=dev, dtype=torch.bfloat16
-                 ).contiguous(memory_format=torch.channels_last
-                 ).requires_grad_(True)
-upsample_nearest2x(xu).float().sum().backward()
+# --- upsample ---
+xu_flat = cl(torch.randn(64, 16, 64, 96, device=dev, dtype=torch.bfloat16)
+             ).permute(0, 2, 3, 1).reshape(-1)
+ext.upsample2x_fwd(xu_flat, 64, 64, 96, 16)
 
 torch.cuda.synchronize()
 print("pmc targets done")

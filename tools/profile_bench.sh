@@ -54,10 +54,10 @@ else
   # PMC collection serializes dispatch-by-dispatch: run the MINIMAL
   # per-kernel target set (tools/pmc_targets.py), never the full bench.
   PMC_TARGET="${PMC_TARGET:-$ROOT/tools/pmc_targets.py}"
-  rocprofv3 --pmc SQ_VALU_MFMA_BUSY_CYCLES SQ_WAVE_CYCLES \
+  timeout 250 rocprofv3 --pmc SQ_VALU_MFMA_BUSY_CYCLES SQ_WAVE_CYCLES \
       SQ_LDS_BANK_CONFLICT -d "$WORK" -o pmc --output-format csv \
     -- python "$PMC_TARGET" > "$WORK/bench.log" 2>&1 || true
-  rocprofv3 --pmc FETCH_SIZE WRITE_SIZE -d "$WORK" -o pmc2 \
+  timeout 250 rocprofv3 --pmc FETCH_SIZE WRITE_SIZE -d "$WORK" -o pmc2 \
       --output-format csv \
     -- python "$PMC_TARGET" > "$WORK/bench2.log" 2>&1 || true
   python3 - "$WORK" "$OUT/${STAMP}_pmc_summary.md" <<'EOF'

@@ -350,7 +350,8 @@ class SynthesisTask:
     # ------------------------------------------------------------------
     def loss_fcn_per_scale(self, scale: int, mpi_packed: torch.Tensor,
                            disparity: torch.Tensor, scale_factor=None,
-                           is_val: bool = False) -> Tuple[dict, dict, torch.Tensor]:
+                           is_val: bool = False,
+                           monitors: bool = True) -> Tuple[dict, dict, torch.Tensor]:
         cfg = self.config
         if scale == 0:
             src_scaled, tgt_scaled = self.src_imgs, self.tgt_imgs
@@ -399,11 +400,20 @@ class SynthesisTask:
         gmin = cfg["loss.smoothness_gmin"]
         grad_ratio = cfg.get("loss.smoothness_grad_ratio", 0.1)
 
+        # The src-view L1/SSIM/smooth-v1 terms are MONITORS (no_grad, ref
+        # synthesis_task.py:301-306): only computed on steps whose values
+        # are read (logging/eval) — the hot loop skips ~3 loss pipelines
+        # x 4 scales per step.
         with torch.no_grad():
-            loss_rgb_src = torch.mean(torch.abs(src_imgs_syn - src_scaled))
-            loss_ssim_src = 1.0 - ssim(src_imgs_syn, src_scaled)
-            loss_smooth_src = edge_aware_loss(src_scaled, src_disparity_syn,
-                                              gmin=gmin, grad_ratio=grad_ratio)
+            if monitors:
+                loss_rgb_src = torch.mean(torch.abs(src_imgs_syn - src_scaled))
+                loss_ssim_src = 1.0 - ssim(src_imgs_syn, src_scaled)
+                loss_smooth_src = edge_aware_loss(
+                    src_scaled, src_disparity_syn,
+                    gmin=gmin, grad_ratio=grad_ratio)
+            else:
+                zero = torch.zeros((), device=self.device)
+                loss_rgb_src = loss_ssim_src = loss_smooth_src = zero
 
         src_pt3d_disp_syn_scaled = src_pt3d_disp_syn / scale_factor.view(B, 1, 1)
         loss_disp_pt3dsrc = disp_lambda * torch.mean(torch.abs(
@@ -421,8 +431,17 @@ class SynthesisTask:
                               cfg["mpi.valid_mask_threshold"]).to(torch.float32)
         loss_rgb_tgt = (torch.abs(tgt_imgs_syn - tgt_scaled) * valid_mask).mean()
 
-        loss_smooth_tgt = lam_v1 * edge_aware_loss(tgt_scaled, tgt_disparity_syn,
-                                                   gmin=gmin, grad_ratio=grad_ratio)
+        if lam_v1 != 0.0:
+            loss_smooth_tgt = lam_v1 * edge_aware_loss(
+                tgt_scaled, tgt_disparity_syn, gmin=gmin,
+                grad_ratio=grad_ratio)
+        elif monitors:
+            with torch.no_grad():
+                loss_smooth_tgt = edge_aware_loss(
+                    tgt_scaled, tgt_disparity_syn, gmin=gmin,
+                    grad_ratio=grad_ratio) * 0.0
+        else:
+            loss_smooth_tgt = torch.zeros((), device=self.device)
         loss_smooth_tgt_v2 = lam_v2 * edge_aware_loss_v2(tgt_scaled, tgt_disparity_syn)
         loss_smooth_src_v2 = lam_v2 * edge_aware_loss_v2(src_scaled, src_disparity_syn)
         loss_ssim_tgt = 1.0 - ssim(tgt_imgs_syn, tgt_scaled)
@@ -432,7 +451,8 @@ class SynthesisTask:
                 lpips_tgt = self.lpips_model(tgt_imgs_syn, tgt_scaled).mean()
             else:
                 lpips_tgt = torch.tensor(0.0, device=self.device)
-            psnr_tgt = psnr(tgt_imgs_syn, tgt_scaled)
+            psnr_tgt = psnr(tgt_imgs_syn, tgt_scaled) if monitors else \
+                torch.zeros((), device=self.device)
 
         loss = (loss_disp_pt3dtgt + loss_disp_pt3dsrc
                 + loss_rgb_tgt + loss_ssim_tgt
@@ -459,7 +479,7 @@ class SynthesisTask:
                     "src_imgs_syn": src_imgs_syn}
         return loss_dict, vis_dict, scale_factor
 
-    def loss_fcn(self, is_val: bool) -> Tuple[dict, dict]:
+    def loss_fcn(self, is_val: bool, monitors: bool = True) -> Tuple[dict, dict]:
         endpoints = self.network_forward()
         mpis = endpoints["mpi_all_src_list"]
         disparity = endpoints["disparity_all_src"]
@@ -468,7 +488,8 @@ class SynthesisTask:
         loss_dicts, vis_dicts = [], []
         for scale in range(4):
             ld, vd, scale_factor = self.loss_fcn_per_scale(
-                scale, mpis[scale], disparity, scale_factor, is_val=is_val)
+                scale, mpis[scale], disparity, scale_factor, is_val=is_val,
+                monitors=monitors)
             loss_dicts.append(ld)
             vis_dicts.append(vd)
 
@@ -496,7 +517,11 @@ class SynthesisTask:
         mark("t0")
         self.set_data(items)
         mark("set_data")
-        loss_dict, _ = self.loss_fcn(is_val=False)
+        # monitor-only loss terms are read on logging steps only
+        log_every = int(self.config.get("training.log_interval", 10))
+        want_monitors = (self.global_step % log_every == 0) or \
+            bool(self.config.get("training.always_monitors", False))
+        loss_dict, _ = self.loss_fcn(is_val=False, monitors=want_monitors)
         mark("forward")
         if self.grad_engine is not None:
             self.grad_engine.zero_grad()
@@ -591,8 +616,10 @@ class SynthesisTask:
             self.global_step += 1
             loss_dict = self.train_step(items)
 
-            if step % int(cfg.get("training.log_interval", 10)) == 0 and \
-                    self.state.is_rank0:
+            # gate on global_step (the same counter train_step uses to
+            # decide whether the monitor losses were computed this step)
+            if self.global_step % int(cfg.get("training.log_interval", 10)) \
+                    == 0 and self.state.is_rank0:
                 self._log_training(epoch, step, len(train_loader), loss_dict)
 
             ckpt_every = int(cfg.get("training.checkpoint_interval", 5000))

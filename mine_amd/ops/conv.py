@@ -159,6 +159,11 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
               and x.is_contiguous(memory_format=torch.channels_last))
     if usable:
         return _Conv3x3ReflFn.apply(x, w, bias)
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        # wide-channel / narrow-image shapes (the C=128 decoder blocks):
+        # the general igemm family, reflect pad in the coordinate map
+        from mine_amd.ops.conv_general import conv2d_mfma
+        return conv2d_mfma(x, w, bias, reflect=True)
     from mine_amd.ops.pad import reflection_pad2d
     return F.conv2d(reflection_pad2d(x, 1), w.to(x.dtype),
                     bias.to(x.dtype) if bias is not None else None)

@@ -123,6 +123,16 @@ class _ConvIgemmFn(torch.autograd.Function):
         x_flat = x.permute(0, 2, 3, 1).reshape(-1)
 
         gx = gw = gb = None
+        Kp = (K + 7) & ~7
+        if K % 8:
+            # the data-grad reads gy in 8-channel fragments: zero-pad
+            # the 4-channel dispconv gradient once here
+            gy_p = torch.cat(
+                (gy, gy.new_zeros(B, Kp - K, P, Q)),
+                1).contiguous(memory_format=torch.channels_last)
+            gy_flat_p = gy_p.permute(0, 2, 3, 1).reshape(-1)
+        else:
+            gy_flat_p = gy_flat
         if ctx.needs_input_grad[1]:
             # wrw maps OUT pixels through the forward tap coordinates
             dw = ext.conv_igemm_wrw(
@@ -134,20 +144,19 @@ class _ConvIgemmFn(torch.autograd.Function):
         if ctx.has_bias:
             gb = gy.float().sum((0, 2, 3)).to(w.dtype)
         if ctx.needs_input_grad[0]:
-            Kp = (K + 7) & ~7
             wtp = pack_weights_general(w, trans=True)
             empty = torch.empty(0, device=x.device, dtype=torch.float32)
             if reflect:
                 # grad to the (virtually) padded input, then reflect-fold
                 Hp, Wp_ = Hs + 2 * pad, Ws + 2 * pad
                 gxp = ext.conv_igemm_fwd(
-                    gy_flat, wtp, empty, B * Hp * Wp_, Hp, Wp_, Cp,
+                    gy_flat_p, wtp, empty, B * Hp * Wp_, Hp, Wp_, Cp,
                     P, Q, Kp, R, S, 1, -1, 0, 1, 0)
                 gx = ext.reflect_pad_bwd(gxp, B, Hs, Ws, Cp, pad)
                 gx = gx.view(B, Hs, Ws, Cp).permute(0, 3, 1, 2)
             else:
                 gxf = ext.conv_igemm_fwd(
-                    gy_flat, wtp, empty, B * Hs * Ws, Hs, Ws, Cp,
+                    gy_flat_p, wtp, empty, B * Hs * Ws, Hs, Ws, Cp,
                     P, Q, Kp, R, S, 1, -1, pad, stride, 0)
                 gx = gxf.view(B, Hs, Ws, Cp).permute(0, 3, 1, 2)
             if Cp != C:
@@ -164,7 +173,6 @@ def conv2d_mfma(x: torch.Tensor, w: torch.Tensor,
     K, C, R, S = w.shape
     usable = (x.is_cuda and x.dtype == torch.bfloat16
               and (R, S) in ((1, 1), (3, 3), (7, 7))
-              and K % 8 == 0  # gy rows feed 8-wide loads in the data-grad
               and w.is_contiguous()
               and x.is_contiguous(memory_format=torch.channels_last)
               and (not reflect or stride == 1))

@@ -31,6 +31,12 @@ void mine_conv_igemm_fwd(const void*, const void*, const float*, void*,
 void mine_conv_igemm_wrw(const void*, const void*, float*, int64_t, int, int,
                          int, int, int, int, int, int, int, int, int, int,
                          int, hipStream_t);
+void mine_eav2_fwd(const float*, const float*, const float*, float*, int,
+                   int, int, int64_t, int64_t, int64_t, int64_t,
+                   hipStream_t);
+void mine_eav2_bwd(const float*, const float*, const float*, float*, float*,
+                   const float*, float*, int, int, int, int64_t, int64_t,
+                   int64_t, int64_t, hipStream_t);
 void mine_pack_gather(const void*, const int*, void*, int64_t, int,
                       hipStream_t);
 void mine_upsample2x_fwd(const void*, void*, int64_t, int64_t, int64_t,
@@ -190,6 +196,35 @@ at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor hfwd,
 
 // --------------------------------------------------------------------------
 // general igemm conv (encoder/neck/base shapes; igemm_kernels.hip)
+
+// fused edge-aware smoothness v2 (loss_kernels.hip)
+at::Tensor eav2_fwd(at::Tensor disp, at::Tensor img, at::Tensor mean_d) {
+  TORCH_CHECK(disp.is_cuda() && disp.is_contiguous() &&
+              disp.scalar_type() == at::kFloat &&
+              img.scalar_type() == at::kFloat);
+  const int B = disp.size(0), H = disp.size(2), W = disp.size(3);
+  auto out = at::zeros({2}, disp.options());
+  mine_eav2_fwd(disp.data_ptr<float>(), img.data_ptr<float>(),
+                mean_d.data_ptr<float>(), out.data_ptr<float>(), B, H, W,
+                img.stride(0), img.stride(1), img.stride(2), img.stride(3),
+                stream());
+  return out;
+}
+
+at::Tensor eav2_bwd(at::Tensor disp, at::Tensor img, at::Tensor mean_d,
+                    at::Tensor gl) {
+  const int B = disp.size(0), H = disp.size(2), W = disp.size(3);
+  auto g_d = at::empty({B, 1, H, W}, disp.options());
+  auto Tb = at::zeros({B}, disp.options());
+  auto grad = at::empty_like(disp);
+  mine_eav2_bwd(disp.data_ptr<float>(), img.data_ptr<float>(),
+                mean_d.data_ptr<float>(), g_d.data_ptr<float>(),
+                Tb.data_ptr<float>(), gl.data_ptr<float>(),
+                grad.data_ptr<float>(), B, H, W,
+                img.stride(0), img.stride(1), img.stride(2), img.stride(3),
+                stream());
+  return grad;
+}
 
 // one-launch fragment pack: bf16 gather through a device LUT (neg -> 0)
 at::Tensor pack_gather(at::Tensor w, at::Tensor lut) {
@@ -555,6 +590,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("tgt_composite_bwd", &tgt_composite_bwd);
   mod.def("ssim_fwd", &ssim_fwd);
   mod.def("ssim_bwd", &ssim_bwd);
+  mod.def("eav2_fwd", &eav2_fwd, "fused edge-aware smoothness v2 fwd");
+  mod.def("eav2_bwd", &eav2_bwd, "fused edge-aware smoothness v2 bwd");
   mod.def("pack_gather", &pack_gather,
           "one-launch fragment pack through a device LUT");
   mod.def("conv_igemm_fwd", &conv_igemm_fwd,

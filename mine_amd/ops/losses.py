@@ -58,8 +58,40 @@ def edge_aware_loss(img: torch.Tensor, disp: torch.Tensor,
     return (loss_x + loss_y).mean()
 
 
+class _EdgeAwareV2Fn(torch.autograd.Function):
+    """Fused edge-aware smoothness v2 (loss_kernels.hip): one reduction
+    kernel forward, gather + finish backward; grad flows to disp only
+    (images carry no gradient in the training losses)."""
+
+    @staticmethod
+    def forward(ctx, img, disp):
+        from mine_amd.ops.backend import get_extension
+        ext = get_extension(required=True)
+        mean_d = disp.detach().mean((1, 2, 3))
+        out = ext.eav2_fwd(disp, img, mean_d)
+        ctx.save_for_backward(img, disp, mean_d)
+        return out[0] + out[1]
+
+    @staticmethod
+    def backward(ctx, gl):
+        from mine_amd.ops.backend import get_extension
+        ext = get_extension(required=True)
+        img, disp, mean_d = ctx.saved_tensors
+        grad = ext.eav2_bwd(disp, img, mean_d, gl.reshape(1).contiguous())
+        return None, grad
+
+
 def edge_aware_loss_v2(img: torch.Tensor, disp: torch.Tensor) -> torch.Tensor:
-    """Mean-normalized edge-aware smoothness (ref network/layers.py:83-99)."""
+    """Mean-normalized edge-aware smoothness (ref network/layers.py:83-99).
+
+    GPU fp32 runs the fused HIP kernels; the eager form is the CPU path
+    and the oracle."""
+    if (disp.is_cuda and disp.dtype == torch.float32
+            and img.dtype == torch.float32 and disp.shape[1] == 1
+            and img.shape[1] == 3 and not img.requires_grad
+            and disp.is_contiguous()):
+        return _EdgeAwareV2Fn.apply(img, disp)
+
     mean_disp = disp.mean(2, True).mean(3, True)
     d = disp / (mean_disp + 1e-7)
 

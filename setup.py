@@ -29,6 +29,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "wrw_kernels.hip"),
         os.path.join(CSRC, "resample_kernels.hip"),
         os.path.join(CSRC, "igemm_kernels.hip"),
+        os.path.join(CSRC, "loss_kernels.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

@@ -623,3 +623,27 @@ def test_conv_igemm_matches_torch(shape):
                                    rtol=5e-2, atol=1e-1)
     torch.testing.assert_close(w.grad.cpu(), wq.grad, rtol=5e-2, atol=5e-1)
     torch.testing.assert_close(b.grad.cpu(), bq.grad, rtol=5e-2, atol=5e-1)
+
+
+def test_edge_aware_v2_fused_matches_eager():
+    from mine_amd.ops.losses import edge_aware_loss_v2, _EdgeAwareV2Fn
+
+    g = torch.Generator().manual_seed(21)
+    for (B, H, W) in [(4, 64, 96), (2, 33, 47)]:
+        img0 = torch.rand(B, 3, H, W, generator=g)
+        disp0 = (torch.rand(B, 1, H, W, generator=g) * 2 + 0.1)
+
+        d_gpu = disp0.cuda().requires_grad_(True)
+        img_gpu = img0.cuda()
+        loss = edge_aware_loss_v2(img_gpu, d_gpu)
+        assert loss.grad_fn is not None and \
+            "EdgeAwareV2" in type(loss.grad_fn).__name__
+        (loss * 3.0).backward()
+
+        d_ref = disp0.clone().requires_grad_(True)
+        loss_ref = edge_aware_loss_v2(img0, d_ref)
+        (loss_ref * 3.0).backward()
+
+        torch.testing.assert_close(loss.cpu(), loss_ref, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(d_gpu.grad.cpu(), d_ref.grad,
+                                   rtol=1e-3, atol=1e-5)

@@ -151,17 +151,18 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
     # 4.2x vs pad+MIOpen at 256x16x256x384); for C>64 or narrow images
     # MIOpen's tuned igemm is better and the pad recompute in backward
     # is not paid back.
-    # (C % 16: the wrw kernel's c-groups are 16-wide; every decoder
-    # shape has C in {16, 32, 64})
+    # (C % 16: the wrw kernel's c-groups are 16-wide; C <= 128 keeps the
+    # 6-row LDS stage under the 160 KiB budget)
     usable = (x.is_cuda and x.dtype == torch.bfloat16
-              and x.shape[1] % 16 == 0 and x.shape[1] <= 64
+              and x.shape[1] % 16 == 0 and x.shape[1] <= 128
               and x.shape[-1] >= 48 and x.shape[-2] >= 8
               and x.is_contiguous(memory_format=torch.channels_last))
     if usable:
         return _Conv3x3ReflFn.apply(x, w, bias)
-    if x.is_cuda and x.dtype == torch.bfloat16:
-        # wide-channel / narrow-image shapes (the C=128 decoder blocks):
-        # the general igemm family, reflect pad in the coordinate map
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.numel() * 2 <= 1 << 25:
+        # small (L2/L3-resident) narrow-image shapes, e.g. the 256-ch
+        # decoder block at H/16: the general igemm family (its direct
+        # loads re-read x per tap, so gate on cache residency)
         from mine_amd.ops.conv_general import conv2d_mfma
         return conv2d_mfma(x, w, bias, reflect=True)
     from mine_amd.ops.pad import reflection_pad2d

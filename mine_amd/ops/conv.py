@@ -151,10 +151,12 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
     # 4.2x vs pad+MIOpen at 256x16x256x384); for C>64 or narrow images
     # MIOpen's tuned igemm is better and the pad recompute in backward
     # is not paid back.
-    # (C % 16: the wrw kernel's c-groups are 16-wide; C <= 128 keeps the
-    # 6-row LDS stage under the 160 KiB budget)
+    # (C % 16: the wrw kernel's c-groups are 16-wide. C <= 64: at C=128
+    # the 101 KiB LDS stage drops occupancy to one workgroup per CU and
+    # measured slower than the library igemm — the C=128 wide blocks
+    # stay on the library path until a bandwidth-tiled variant exists.)
     usable = (x.is_cuda and x.dtype == torch.bfloat16
-              and x.shape[1] % 16 == 0 and x.shape[1] <= 128
+              and x.shape[1] % 16 == 0 and x.shape[1] <= 64
               and x.shape[-1] >= 48 and x.shape[-2] >= 8
               and x.is_contiguous(memory_format=torch.channels_last))
     if usable:

@@ -166,7 +166,8 @@ class SynthesisTask:
             self.grad_engine = GradAllReduceEngine(
                 [self.backbone, self.decoder],
                 bucket_mb=float(config.get("training.grad_bucket_mb", 25)),
-                allreduce_dtype=ar_dtype)
+                allreduce_dtype=ar_dtype,
+                timing=bool(config.get("training.comm_timing", False)))
             self.lr_scheduler = torch.optim.lr_scheduler.MultiStepLR(
                 self.optimizer, config["lr.decay_steps"],
                 gamma=config["lr.decay_gamma"])

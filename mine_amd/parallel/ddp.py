@@ -48,7 +48,8 @@ def init_distributed(backend: Optional[str] = None) -> tuple:
 
 
 class _Bucket:
-    __slots__ = ("flat", "comm", "params", "ready", "handle", "offsets")
+    __slots__ = ("flat", "comm", "params", "ready", "handle", "offsets",
+                 "ev_launch", "ev_done")
 
     def __init__(self, flat: torch.Tensor, params: List[torch.nn.Parameter],
                  offsets: List[int], comm: Optional[torch.Tensor] = None):
@@ -58,6 +59,8 @@ class _Bucket:
         self.offsets = offsets
         self.ready = 0
         self.handle = None
+        self.ev_launch = None  # comm-timing events (timing mode only)
+        self.ev_done = None
 
 
 class GradAllReduceEngine:
@@ -74,15 +77,21 @@ class GradAllReduceEngine:
                  bucket_mb: float = 25.0,
                  process_group: Optional[dist.ProcessGroup] = None,
                  broadcast_params: bool = True,
-                 allreduce_dtype: Optional[torch.dtype] = None):
+                 allreduce_dtype: Optional[torch.dtype] = None,
+                 timing: bool = False):
         """allreduce_dtype: wire dtype for the gradient collectives
         (e.g. torch.bfloat16 halves the xGMI bytes; grads still
         ACCUMULATE in the fp32 buckets — only the reduction itself is
-        compressed). None = reduce the fp32 buckets directly."""
+        compressed). None = reduce the fp32 buckets directly.
+        timing: record per-bucket launch->completion times (CUDA events;
+        read with pop_bucket_times()) so the first scale run on real
+        hardware yields a diagnosable overlap picture."""
         self.pg = process_group
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.enabled = self.world_size > 1
         self.allreduce_dtype = allreduce_dtype
+        self.timing = bool(timing) and torch.cuda.is_available()
+        self._bucket_times: List[List[float]] = []
 
         params: List[torch.nn.Parameter] = []
         for m in modules:
@@ -154,6 +163,9 @@ class GradAllReduceEngine:
 
     # ------------------------------------------------------------------
     def _launch(self, b: "_Bucket") -> None:
+        if self.timing:
+            b.ev_launch = torch.cuda.Event(enable_timing=True)
+            b.ev_launch.record()
         if b.comm is not None:
             b.comm.copy_(b.flat)
             b.handle = dist.all_reduce(b.comm, op=dist.ReduceOp.SUM,
@@ -161,6 +173,9 @@ class GradAllReduceEngine:
         else:
             b.handle = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                        group=self.pg, async_op=True)
+        if self.timing:
+            b.ev_done = torch.cuda.Event(enable_timing=True)
+            b.ev_done.record()
 
     def _on_grad_ready(self, p: torch.nn.Parameter) -> None:
         b, _ = self._param_bucket[id(p)]
@@ -191,6 +206,18 @@ class GradAllReduceEngine:
             b.flat.mul_(inv)
             b.handle = None
             b.ready = 0
+        if self.timing:
+            torch.cuda.synchronize()
+            self._bucket_times.append(
+                [b.ev_launch.elapsed_time(b.ev_done)
+                 if b.ev_launch is not None else -1.0
+                 for b in self.buckets])
+
+    def pop_bucket_times(self) -> List[List[float]]:
+        """Per-step per-bucket all-reduce stream times (ms) recorded in
+        timing mode; cleared on read."""
+        t, self._bucket_times = self._bucket_times, []
+        return t
 
     def detach(self) -> None:
         for h in self._hooks:

@@ -354,22 +354,27 @@ class SynthesisTask:
                            is_val: bool = False,
                            monitors: bool = True) -> Tuple[dict, dict, torch.Tensor]:
         cfg = self.config
-        if scale == 0:
+        # pyramid sized from the decoder's OWN per-scale output: at
+        # non-power-of-two resolutions (LLFF 504x378) H//2^s and the
+        # up-stage chain disagree by one pixel (the reference crashes
+        # there); intrinsics scale per-axis by the actual ratio.
+        Hs_m, Ws_m = int(mpi_packed.shape[2]), int(mpi_packed.shape[3])
+        if scale == 0 and (Hs_m, Ws_m) == tuple(self.src_imgs.shape[-2:]):
             src_scaled, tgt_scaled = self.src_imgs, self.tgt_imgs
         else:
-            size = (self.src_imgs.shape[-2] // 2 ** scale,
-                    self.src_imgs.shape[-1] // 2 ** scale)
-            src_scaled = F.interpolate(self.src_imgs, size=size, mode="nearest")
-            tgt_scaled = F.interpolate(self.tgt_imgs, size=size, mode="nearest")
+            src_scaled = F.interpolate(self.src_imgs, size=(Hs_m, Ws_m),
+                                       mode="nearest")
+            tgt_scaled = F.interpolate(self.tgt_imgs, size=(Hs_m, Ws_m),
+                                       mode="nearest")
         B = src_scaled.shape[0]
 
-        K_src_scaled = self.K_src / (2 ** scale)
-        K_src_scaled[:, 2, 2] = 1.0
-        K_tgt_scaled = self.K_tgt / (2 ** scale)
-        K_tgt_scaled[:, 2, 2] = 1.0
+        sy = Hs_m / self.src_imgs.shape[-2]
+        sx = Ws_m / self.src_imgs.shape[-1]
+        sc = torch.tensor([[sx], [sy], [1.0]], dtype=torch.float32,
+                          device=self.device)
+        K_src_scaled = self.K_src * sc
+        K_tgt_scaled = self.K_tgt * sc
         K_src_scaled_inv = inverse_3x3(K_src_scaled)
-
-        assert mpi_packed.shape[2] == src_scaled.shape[-2]
 
         # ---- fused src composite + RGB blending -------------------------
         src_imgs_syn, src_depth_syn, mpi_blend = render_src_view(

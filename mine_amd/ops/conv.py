@@ -151,6 +151,11 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
     # 4.2x vs pad+MIOpen at 256x16x256x384); for C>64 or narrow images
     # MIOpen's tuned igemm is better and the pad recompute in backward
     # is not paid back.
+    if x.is_cuda and x.dtype == torch.float16:
+        # fp16 configs (Flowers, driver config 5): the MFMA kernels are
+        # bf16; one cast pass each way beats the library fp16 path and
+        # its per-shape find cost. Accumulation is fp32 either way.
+        return conv3x3_reflect(x.to(torch.bfloat16), w, bias).to(torch.float16)
     # (C % 16: the wrw kernel's c-groups are 16-wide. C <= 64: at C=128
     # the 101 KiB LDS stage drops occupancy to one workgroup per CU and
     # measured slower than the library igemm — the C=128 wide blocks

@@ -171,6 +171,11 @@ def conv2d_mfma(x: torch.Tensor, w: torch.Tensor,
     fast-path conditions don't hold. reflect=True means
     ReflectionPad((R-1)/2) + unpadded conv (stride must be 1)."""
     K, C, R, S = w.shape
+    if x.is_cuda and x.dtype == torch.float16 and (R, S) in (
+            (1, 1), (3, 3), (7, 7)) and (not reflect or stride == 1):
+        # fp16 configs: cast through the bf16 MFMA path (see conv.py)
+        return conv2d_mfma(x.to(torch.bfloat16), w, bias, stride, padding,
+                           reflect).to(torch.float16)
     usable = (x.is_cuda and x.dtype == torch.bfloat16
               and (R, S) in ((1, 1), (3, 3), (7, 7))
               and w.is_contiguous()

@@ -285,3 +285,14 @@ def test_get_dataset_fails_loudly_without_data():
                    "data.allow_synthetic_fallback": True})
     ds = get_dataset(cfg2)
     assert len(ds) > 0
+
+
+def test_odd_resolution_train_step():
+    """Non-power-of-two resolutions (the LLFF 504x378 config): the loss
+    pyramid sizes itself from the decoder's per-scale outputs and the
+    intrinsics scale per-axis (the reference crashes on such shapes)."""
+    cfg = _cfg(**{"data.img_h": 38, "data.img_w": 52,
+                  "data.synthetic_length": 2})
+    task = SynthesisTask(cfg, device="cpu")
+    loss = task.train_step(_items(cfg))
+    assert torch.isfinite(loss["loss"])

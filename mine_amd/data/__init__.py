@@ -25,6 +25,14 @@ def get_dataset(config, logger=None, is_validation: bool = False):
                            supervision_count=config["data.num_tgt_views"],
                            visible_points_count=config["data.visible_point_count"],
                            img_pre_downsample_ratio=config["data.img_pre_downsample_ratio"])
+    if name == "flowers":
+        import os
+        root = config["data.training_set_path"]
+        if isinstance(root, str) and \
+                os.path.exists(os.path.join(root, "cam_params.txt")):
+            from mine_amd.data.flowers import FlowersDataset
+            return FlowersDataset(config, logger, root=root,
+                                  is_validation=is_validation)
     if name != "synthetic" and not bool(
             config.get("data.allow_synthetic_fallback", False)):
         raise NotImplementedError(

@@ -431,7 +431,8 @@ def test_flowers_pipeline(tmp_path):
     assert src["img"].shape == (3, 48, 64)
     assert tgts[0]["G_src_tgt"].shape == (4, 4)
 
-    # one full CPU train step on flowers items (scale factor == 1 path)
+    # one full CPU train step on flowers items (scale factor == 1 path;
+    # batch 2: batch-1 BN over a 1x1 deep tap is degenerate)
     task = SynthesisTask(cfg, device="cpu")
-    loss = task.train_step(collate_src_tgt([ds[0]]))
+    loss = task.train_step(collate_src_tgt([ds[0], ds[0]]))
     assert torch.isfinite(loss["loss"])

@@ -13,18 +13,19 @@
 extern "C" {
 void mine_src_composite_fwd(const float*, const float*, const float*,
                             const float*, float*, float*, float*, int, int,
-                            int, int, int, hipStream_t);
+                            int, int, int, int, hipStream_t);
 void mine_src_composite_bwd(const float*, const float*, const float*,
                             const float*, const float*, const float*,
                             const float*, float*, int, int, int, int, int,
-                            hipStream_t);
+                            int, hipStream_t);
 void mine_tgt_composite_fwd(const float*, const float*, const float*,
                             const float*, const float*, float*, float*,
-                            float*, int, int, int, int, int, hipStream_t);
+                            float*, int, int, int, int, int, int,
+                            hipStream_t);
 void mine_tgt_composite_bwd(const float*, const float*, const float*,
                             const float*, const float*, const float*,
                             const float*, const float*, float*, float*,
-                            int, int, int, int, int, int, hipStream_t);
+                            int, int, int, int, int, int, int, hipStream_t);
 void mine_conv_igemm_fwd(const void*, const void*, const float*, void*,
                          float*, int64_t, int, int, int, int, int, int, int,
                          int, int, int, int, int, int, int, hipStream_t);
@@ -111,7 +112,7 @@ hipStream_t stream() {
 
 std::vector<at::Tensor> src_composite_fwd(at::Tensor mpi, at::Tensor depths,
                                           at::Tensor kinv, at::Tensor img,
-                                          bool bg_inf) {
+                                          bool bg_inf, bool alpha) {
   CHECK_IN(mpi); CHECK_IN(depths); CHECK_IN(kinv);
   TORCH_CHECK(mpi.dim() == 5 && mpi.size(4) == 4, "mpi must be (B,S,H,W,4)");
   const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
@@ -129,14 +130,15 @@ std::vector<at::Tensor> src_composite_fwd(at::Tensor mpi, at::Tensor depths,
                          kinv.data_ptr<float>(), optr(img),
                          rgb.data_ptr<float>(), depth.data_ptr<float>(),
                          blend ? mpi_blend.data_ptr<float>() : nullptr,
-                         B, S, H, W, bg_inf ? 1 : 0, stream());
+                         B, S, H, W, bg_inf ? 1 : 0, alpha ? 1 : 0,
+                         stream());
   return {rgb, depth, mpi_blend};
 }
 
 at::Tensor src_composite_bwd(at::Tensor mpi, at::Tensor depths,
                              at::Tensor kinv, at::Tensor img, bool bg_inf,
-                             at::Tensor g_rgb, at::Tensor g_depth,
-                             at::Tensor g_blend) {
+                             bool alpha, at::Tensor g_rgb,
+                             at::Tensor g_depth, at::Tensor g_blend) {
   CHECK_IN(mpi);
   const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
   auto grad_mpi = at::empty_like(mpi);
@@ -144,13 +146,14 @@ at::Tensor src_composite_bwd(at::Tensor mpi, at::Tensor depths,
                          kinv.data_ptr<float>(), optr(img), optr(g_rgb),
                          optr(g_depth), optr(g_blend),
                          grad_mpi.data_ptr<float>(), B, S, H, W,
-                         bg_inf ? 1 : 0, stream());
+                         bg_inf ? 1 : 0, alpha ? 1 : 0, stream());
   return grad_mpi;
 }
 
 std::vector<at::Tensor> tgt_composite_fwd(at::Tensor mpi, at::Tensor hinv,
                                           at::Tensor m, at::Tensor tvec,
-                                          at::Tensor depths, bool bg_inf) {
+                                          at::Tensor depths, bool bg_inf,
+                                          bool alpha) {
   CHECK_IN(mpi); CHECK_IN(hinv); CHECK_IN(m); CHECK_IN(tvec); CHECK_IN(depths);
   TORCH_CHECK(mpi.dim() == 5 && mpi.size(4) == 4, "mpi must be (B,S,H,W,4)");
   const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
@@ -163,7 +166,8 @@ std::vector<at::Tensor> tgt_composite_fwd(at::Tensor mpi, at::Tensor hinv,
                          m.data_ptr<float>(), tvec.data_ptr<float>(),
                          depths.data_ptr<float>(), rgb.data_ptr<float>(),
                          depth.data_ptr<float>(), mask.data_ptr<float>(),
-                         B, S, H, W, bg_inf ? 1 : 0, stream());
+                         B, S, H, W, bg_inf ? 1 : 0, alpha ? 1 : 0,
+                         stream());
   return {rgb, depth, mask};
 }
 
@@ -173,8 +177,8 @@ std::vector<at::Tensor> tgt_composite_fwd(at::Tensor mpi, at::Tensor hinv,
 // A/B tests and as a fallback).
 at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor hfwd,
                              at::Tensor m, at::Tensor tvec, at::Tensor depths,
-                             bool bg_inf, at::Tensor g_rgb, at::Tensor g_depth,
-                             int64_t mode) {
+                             bool bg_inf, bool alpha, at::Tensor g_rgb,
+                             at::Tensor g_depth, int64_t mode) {
   CHECK_IN(mpi);
   const int B = mpi.size(0), S = mpi.size(1), H = mpi.size(2), W = mpi.size(3);
   auto grad_mpi = at::zeros_like(mpi);
@@ -190,7 +194,7 @@ at::Tensor tgt_composite_bwd(at::Tensor mpi, at::Tensor hinv, at::Tensor hfwd,
                          m.data_ptr<float>(), tvec.data_ptr<float>(),
                          depths.data_ptr<float>(), optr(g_rgb), optr(g_depth),
                          grad_mpi.data_ptr<float>(), pay_ptr, B, S, H, W,
-                         bg_inf ? 1 : 0, (int)mode, stream());
+                         bg_inf ? 1 : 0, (int)mode, alpha ? 1 : 0, stream());
   return grad_mpi;
 }
 

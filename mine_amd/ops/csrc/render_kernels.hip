@@ -58,7 +58,12 @@ DEV float clampf(float v, float lo, float hi) {
 //   R = sum w*c ; D = sum(w*z)/(sum w + 1e-5)  (or + 1000*(1-sum w))
 // ref operations/mpi_rendering.py:42-82, synthesis_task.py:267-274.
 
-template <bool BLEND, bool BG_INF>
+// ALPHA: the 4th MPI channel is an alpha in (0,1) (ref
+// operations/mpi_rendering.py:23-39): t = 1-alpha (no exp, no plane
+// distance), the running product has NO epsilon, and the depth is the
+// PLAIN weighted z sum (no normalization, no background term). BLEND
+// is never combined with ALPHA (ref mpi_rendering.py:19).
+template <bool BLEND, bool BG_INF, bool ALPHA = false>
 __global__ void __launch_bounds__(kBlock)
 src_composite_fwd_kernel(const float* __restrict__ mpi,
                          const float* __restrict__ depths,   // (B,S)
@@ -99,7 +104,7 @@ src_composite_fwd_kernel(const float* __restrict__ mpi,
           mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
       const float d = s_depth[s];
       const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
-      const float t = __expf(-px.w * delta);
+      const float t = ALPHA ? (1.0f - px.w) : __expf(-px.w * delta);
       const float w = A * (1.0f - t);
       float3 c = make_float3(px.x, px.y, px.z);
       if (BLEND) {
@@ -112,7 +117,7 @@ src_composite_fwd_kernel(const float* __restrict__ mpi,
       R.x += w * c.x; R.y += w * c.y; R.z += w * c.z;
       Wsum += w;
       Nsum += w * d;  // src-view z of plane s is its depth
-      A *= (t + 1e-6f);
+      A *= ALPHA ? t : (t + 1e-6f);
       if (A < 1e-14f) {  // dead transmittance: remaining planes add ~0
         if (BLEND) {
           // the blended tail is c = A*I + (1-A)*rgb ~= rgb: copy through
@@ -125,8 +130,9 @@ src_composite_fwd_kernel(const float* __restrict__ mpi,
         break;
       }
     }
-    const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
-                           : (Nsum / (Wsum + 1e-5f));
+    const float D = ALPHA ? Nsum
+                          : (BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
+                                    : (Nsum / (Wsum + 1e-5f)));
     rgb_out[((int64_t)b * 3 + 0) * HW + pix] = R.x;
     rgb_out[((int64_t)b * 3 + 1) * HW + pix] = R.y;
     rgb_out[((int64_t)b * 3 + 2) * HW + pix] = R.z;
@@ -150,12 +156,12 @@ struct PlaneTerm {
   float dcx, dcy, dcz;  // dL/dc_s
 };
 
-template <bool BLEND, bool BG_INF>
+template <bool BLEND, bool BG_INF, bool ALPHA = false>
 __device__ __attribute__((noinline)) PlaneTerm
 src_plane_term(float4 px, float Af, float3 I, float3 gR, float gD,
                float d, float delta, float D, float Wp, float3 gC) {
   PlaneTerm r;
-  r.t = __expf(-px.w * delta);
+  r.t = ALPHA ? (1.0f - px.w) : __expf(-px.w * delta);
   const float w = Af * (1.0f - r.t);
   float3 c = make_float3(px.x, px.y, px.z);
   if (BLEND) {
@@ -167,7 +173,7 @@ src_plane_term(float4 px, float Af, float3 I, float3 gR, float gD,
   r.dcy = w * gR.y + gC.y;
   r.dcz = w * gR.z + gC.z;
   r.e = c.x * gR.x + c.y * gR.y + c.z * gR.z +
-        gD * (BG_INF ? (d - 1000.0f) : (d - D) / Wp);
+        gD * (ALPHA ? d : (BG_INF ? (d - 1000.0f) : (d - D) / Wp));
   float dA = (1.0f - r.t) * r.e;
   if (BLEND) {
     dA += (I.x - px.x) * r.dcx + (I.y - px.y) * r.dcy + (I.z - px.z) * r.dcz;
@@ -176,7 +182,7 @@ src_plane_term(float4 px, float Af, float3 I, float3 gR, float gD,
   return r;
 }
 
-template <bool BLEND, bool BG_INF>
+template <bool BLEND, bool BG_INF, bool ALPHA = false>
 __global__ void __launch_bounds__(kBlock)
 src_composite_bwd_kernel(const float* __restrict__ mpi,
                          const float* __restrict__ depths,
@@ -225,14 +231,16 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
           mpi + mpi_b + ((int64_t)s * HW + pix) * 4);
       const float d = s_depth[s];
       const float delta = (s + 1 < S) ? nu * (s_depth[s + 1] - d) : 1e3f;
-      const float t = __expf(-px.w * delta);
+      const float t = ALPHA ? (1.0f - px.w) : __expf(-px.w * delta);
       const float w = A * (1.0f - t);
       Wsum += w;
       Nsum += w * d;
-      A *= (t + 1e-6f);
+      A *= ALPHA ? t : (t + 1e-6f);
     }
     const float Wp = Wsum + 1e-5f;
-    const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
+    const float D = ALPHA ? Nsum
+                          : (BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
+                                    : (Nsum / Wp));
 
     // ---- pass 2: total cumprod-suffix mass ----
     // fp64 transmittance + accumulators: pass 3 computes
@@ -256,10 +264,10 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
             g_blend + mpi_b + ((int64_t)s * HW + pix) * 4);
         gC = make_float3(gb.x, gb.y, gb.z);
       }
-      const PlaneTerm pt = src_plane_term<BLEND, BG_INF>(
+      const PlaneTerm pt = src_plane_term<BLEND, BG_INF, ALPHA>(
           px, (float)Ad, I, gR, gD, d, delta, D, Wp, gC);
       TotalP += (double)pt.dA * Ad;
-      Ad *= (double)(pt.t + 1e-6f);
+      Ad *= (double)(ALPHA ? pt.t : (pt.t + 1e-6f));
       if (Ad < 1e-14) break;
     }
 
@@ -280,12 +288,13 @@ src_composite_bwd_kernel(const float* __restrict__ mpi,
         gCs = gb.w;
       }
       const float Af = (float)Ad;
-      const PlaneTerm pt = src_plane_term<BLEND, BG_INF>(
+      const PlaneTerm pt = src_plane_term<BLEND, BG_INF, ALPHA>(
           px, Af, I, gR, gD, d, delta, D, Wp, gC);
-      const float u = pt.t + 1e-6f;
+      const float u = ALPHA ? pt.t : (pt.t + 1e-6f);
       prefix += (double)pt.dA * Ad;
       const float dt = -Af * pt.e + (float)((TotalP - prefix) / (double)u);
-      const float dsigma = dt * (-delta * pt.t) + gCs;
+      // alpha mode: d(t)/d(alpha) = -1; sigma mode: d(t)/d(sigma) = -delta*t
+      const float dsigma = (ALPHA ? -dt : dt * (-delta * pt.t)) + gCs;
       float4 g;
       if (BLEND) {
         const float oneA = 1.0f - Af;
@@ -400,7 +409,7 @@ sample_plane(const float* __restrict__ mpi_b, int s, int HW,
   return ps;
 }
 
-template <bool BG_INF>
+template <bool BG_INF, bool ALPHA = false>
 __global__ void __launch_bounds__(kBlock)
 tgt_composite_fwd_kernel(const float* __restrict__ mpi,
                          const float* __restrict__ hinv,    // (B,S,3,3)
@@ -450,12 +459,13 @@ tgt_composite_fwd_kernel(const float* __restrict__ mpi,
       } else {
         delta = 1e3f;
       }
-      const float t = __expf(-cur.rgbs.w * delta);
+      const float t = ALPHA ? (1.0f - cur.rgbs.w)
+                            : __expf(-cur.rgbs.w * delta);
       const float w = A * (1.0f - t);
       R.x += w * cur.rgbs.x; R.y += w * cur.rgbs.y; R.z += w * cur.rgbs.z;
       Wsum += w;
       Nsum += w * cur.v.z;
-      A *= (t + 1e-6f);
+      A *= ALPHA ? t : (t + 1e-6f);
       cur = nxt;
       if (A < 1e-14f) {  // transmittance dead: remaining planes add ~0
         ++s;
@@ -474,8 +484,9 @@ tgt_composite_fwd_kernel(const float* __restrict__ mpi,
       mask += (u > -1.0f && u < (float)W && v > -1.0f && v < (float)H)
                   ? 1.0f : 0.0f;
     }
-    const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
-                           : (Nsum / (Wsum + 1e-5f));
+    const float D = ALPHA ? Nsum
+                          : (BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum))
+                                    : (Nsum / (Wsum + 1e-5f)));
     rgb_out[((int64_t)b * 3 + 0) * HW + pix] = R.x;
     rgb_out[((int64_t)b * 3 + 1) * HW + pix] = R.y;
     rgb_out[((int64_t)b * 3 + 2) * HW + pix] = R.z;
@@ -521,10 +532,11 @@ struct TgtTerm {
   float t, r, cr, cv, cw;
 };
 
+template <bool ALPHA>
 __device__ __attribute__((noinline)) TgtTerm
 tgt_plane_terms(float4 rgbs, float vz, float delta, float3 gR) {
   TgtTerm o;
-  o.t = __expf(-rgbs.w * delta);
+  o.t = ALPHA ? (1.0f - rgbs.w) : __expf(-rgbs.w * delta);
   o.r = rgbs.x * gR.x + rgbs.y * gR.y + rgbs.z * gR.z;
   const float omt = 1.0f - o.t;
   o.cr = omt * o.r;
@@ -540,7 +552,7 @@ tgt_plane_terms(float4 rgbs, float vz, float delta, float3 gR) {
 // per plane) take the atomic path. tgt_gather_kernel then inverts the
 // map per source tile. Decomposition proven against the scatter form
 // in tests/test_kernel_sim.py::test_warp_backward_gather_decomposition.
-template <bool BG_INF, bool GATHER>
+template <bool BG_INF, bool GATHER, bool ALPHA = false>
 __global__ void __launch_bounds__(kBlock)
 tgt_composite_bwd_kernel(const float* __restrict__ mpi,
                          const float* __restrict__ hinv,
@@ -596,23 +608,25 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       } else {
         delta = 1e3f;
       }
-      const TgtTerm tt = tgt_plane_terms(cur.rgbs, cur.v.z, delta, gR);
+      const TgtTerm tt = tgt_plane_terms<ALPHA>(cur.rgbs, cur.v.z, delta,
+                                                gR);
       const float w = A * (1.0f - tt.t);
       Wsum += w;
       Nsum += w * cur.v.z;
       TaR += (double)tt.cr * Ad;
       TaV += (double)tt.cv * Ad;
       TaW += (double)tt.cw * Ad;
-      A *= (tt.t + 1e-6f);
-      Ad *= (double)(tt.t + 1e-6f);
+      A *= ALPHA ? tt.t : (tt.t + 1e-6f);
+      Ad *= (double)(ALPHA ? tt.t : (tt.t + 1e-6f));
       cur = nxt;
       if (Ad < 1e-14) break;  // dead transmittance: tail adds ~0
     }
     const float Wp = Wsum + 1e-5f;
     const float D = BG_INF ? (Nsum + 1000.0f * (1.0f - Wsum)) : (Nsum / Wp);
-    // e = r + qv*vz + qc
-    const float qv = BG_INF ? gD : gD / Wp;
-    const float qc = BG_INF ? (-1000.0f * gD) : (-gD / Wp * D);
+    // e = r + qv*vz + qc (alpha: plain weighted z sum -> qv = gD, qc = 0)
+    const float qv = ALPHA ? gD : (BG_INF ? gD : gD / Wp);
+    const float qc = ALPHA ? 0.0f
+                           : (BG_INF ? (-1000.0f * gD) : (-gD / Wp * D));
 
     // ---- pass B: emit gradients, bilinear scatter ----
     double prR = 0.0, prV = 0.0, prW = 0.0;
@@ -631,9 +645,10 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
       } else {
         delta = 1e3f;
       }
-      const TgtTerm tt = tgt_plane_terms(cur.rgbs, cur.v.z, delta, gR);
+      const TgtTerm tt = tgt_plane_terms<ALPHA>(cur.rgbs, cur.v.z, delta,
+                                                gR);
       const float t = tt.t;
-      const float u = t + 1e-6f;
+      const float u = ALPHA ? t : (t + 1e-6f);
       const float Af = (float)Ad;
       const float w = Af * (1.0f - t);
       prR += (double)tt.cr * Ad;
@@ -645,8 +660,9 @@ tgt_composite_bwd_kernel(const float* __restrict__ mpi,
                             (double)qc * (TaW - prW);
       const float e = tt.r + qv * cur.v.z + qc;
       const float dt = -Af * e + (float)(suffix / (double)u);
-      // culled sigma contributed nothing -> no gradient through it
-      const float dsigma = (cur.v.z < 0.0f) ? 0.0f : dt * (-delta * t);
+      // culled sigma/alpha contributed nothing -> no gradient through it
+      const float dsigma = (cur.v.z < 0.0f)
+          ? 0.0f : (ALPHA ? -dt : dt * (-delta * t));
       const float4 g = make_float4(w * gR.x, w * gR.y, w * gR.z, dsigma);
       if (GATHER && tap.interior) {
         *reinterpret_cast<float4*>(
@@ -800,10 +816,16 @@ extern "C" {
 void mine_src_composite_fwd(const float* mpi, const float* depths,
                             const float* kinv, const float* img,
                             float* rgb_out, float* depth_out, float* mpi_blend,
-                            int B, int S, int H, int W, int bg_inf,
+                            int B, int S, int H, int W, int bg_inf, int alpha,
                             hipStream_t stream) {
   dim3 grid(grid_x(H * W), B);
   const bool blend = img != nullptr;
+  if (alpha) {  // alpha-compositing: no blend, no bg term (ref :19-39)
+    hipLaunchKernelGGL((src_composite_fwd_kernel<false, false, true>), grid,
+                       dim3(kBlock), 0, stream, mpi, depths, kinv, nullptr,
+                       rgb_out, depth_out, mpi_blend, B, S, H, W);
+    return;
+  }
   DISPATCH_BOOL(blend, BLEND, {
     DISPATCH_BOOL(bg_inf, BG, {
       hipLaunchKernelGGL((src_composite_fwd_kernel<BLEND, BG>), grid,
@@ -817,10 +839,16 @@ void mine_src_composite_bwd(const float* mpi, const float* depths,
                             const float* kinv, const float* img,
                             const float* g_rgb, const float* g_depth,
                             const float* g_blend, float* grad_mpi,
-                            int B, int S, int H, int W, int bg_inf,
+                            int B, int S, int H, int W, int bg_inf, int alpha,
                             hipStream_t stream) {
   dim3 grid(grid_x(H * W), B);
   const bool blend = img != nullptr;
+  if (alpha) {
+    hipLaunchKernelGGL((src_composite_bwd_kernel<false, false, true>), grid,
+                       dim3(kBlock), 0, stream, mpi, depths, kinv, nullptr,
+                       g_rgb, g_depth, nullptr, grad_mpi, B, S, H, W);
+    return;
+  }
   DISPATCH_BOOL(blend, BLEND, {
     DISPATCH_BOOL(bg_inf, BG, {
       hipLaunchKernelGGL((src_composite_bwd_kernel<BLEND, BG>), grid,
@@ -834,9 +862,15 @@ void mine_tgt_composite_fwd(const float* mpi, const float* hinv,
                             const float* m_rki, const float* tvec,
                             const float* depths, float* rgb_out,
                             float* depth_out, float* mask_out,
-                            int B, int S, int H, int W, int bg_inf,
+                            int B, int S, int H, int W, int bg_inf, int alpha,
                             hipStream_t stream) {
   dim3 grid(grid_x(H * W), B);
+  if (alpha) {
+    hipLaunchKernelGGL((tgt_composite_fwd_kernel<false, true>), grid,
+                       dim3(kBlock), 0, stream, mpi, hinv, m_rki, tvec,
+                       depths, rgb_out, depth_out, mask_out, B, S, H, W);
+    return;
+  }
   DISPATCH_BOOL(bg_inf, BG, {
     hipLaunchKernelGGL((tgt_composite_fwd_kernel<BG>), grid, dim3(kBlock), 0,
                        stream, mpi, hinv, m_rki, tvec, depths, rgb_out,
@@ -852,26 +886,15 @@ void mine_tgt_composite_bwd(const float* mpi, const float* hinv,
                             const float* g_rgb, const float* g_depth,
                             float* grad_mpi, float* payload,
                             int B, int S, int H, int W, int bg_inf, int mode,
-                            hipStream_t stream) {
+                            int alpha, hipStream_t stream) {
   dim3 grid(grid_x(H * W), B);
-  DISPATCH_BOOL(bg_inf, BG, {
-    if (mode == 1) {
-      hipLaunchKernelGGL((tgt_composite_bwd_kernel<BG, true>), grid,
-                         dim3(kBlock), 0, stream, mpi, hinv, m_rki, tvec,
-                         depths, g_rgb, g_depth, grad_mpi, payload,
-                         B, S, H, W);
-      const int tiles_x = (W + GT - 1) / GT;
-      const int tiles_y = (H + GT - 1) / GT;
-      dim3 ggrid(tiles_x * tiles_y, S, B);
-      hipLaunchKernelGGL(tgt_gather_kernel, ggrid, dim3(kBlock), 0, stream,
-                         payload, hinv, hfwd, grad_mpi, B, S, H, W, tiles_x);
-    } else {
-      hipLaunchKernelGGL((tgt_composite_bwd_kernel<BG, false>), grid,
-                         dim3(kBlock), 0, stream, mpi, hinv, m_rki, tvec,
-                         depths, g_rgb, g_depth, grad_mpi, nullptr,
-                         B, S, H, W);
-    }
-  })
+#define TGT_BWD_BODY(BGv, ALv)                                                if (mode == 1) {                                                              hipLaunchKernelGGL((tgt_composite_bwd_kernel<BGv, true, ALv>), grid,                           dim3(kBlock), 0, stream, mpi, hinv, m_rki, tvec,                            depths, g_rgb, g_depth, grad_mpi, payload,                                  B, S, H, W);                                             const int tiles_x = (W + GT - 1) / GT;                                      const int tiles_y = (H + GT - 1) / GT;                                      dim3 ggrid(tiles_x * tiles_y, S, B);                                        hipLaunchKernelGGL(tgt_gather_kernel, ggrid, dim3(kBlock), 0, stream,                          payload, hinv, hfwd, grad_mpi, B, S, H, W, tiles_x);   } else {                                                                      hipLaunchKernelGGL((tgt_composite_bwd_kernel<BGv, false, ALv>), grid,                          dim3(kBlock), 0, stream, mpi, hinv, m_rki, tvec,                            depths, g_rgb, g_depth, grad_mpi, nullptr,                                  B, S, H, W);                                           }
+  if (alpha) {
+    TGT_BWD_BODY(false, true)
+    return;
+  }
+  DISPATCH_BOOL(bg_inf, BG, { TGT_BWD_BODY(BG, false) })
+#undef TGT_BWD_BODY
 }
 
 }  // extern "C"

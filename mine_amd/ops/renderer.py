@@ -82,15 +82,16 @@ class _SrcCompositeFn(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, mpi, depths, k_inv, src_img, bg_inf, blend):
+    def forward(ctx, mpi, depths, k_inv, src_img, bg_inf, blend, alpha):
         ext = get_extension(required=True)
         rgb_syn, depth_syn, mpi_blend = ext.src_composite_fwd(
             mpi, depths, k_inv,
             src_img if blend else torch.empty(0, device=mpi.device, dtype=mpi.dtype),
-            bool(bg_inf))
+            bool(bg_inf), bool(alpha))
         ctx.save_for_backward(mpi, depths, k_inv, src_img)
         ctx.bg_inf = bool(bg_inf)
         ctx.blend = bool(blend)
+        ctx.alpha = bool(alpha)
         return rgb_syn, depth_syn, mpi_blend
 
     @staticmethod
@@ -101,11 +102,11 @@ class _SrcCompositeFn(torch.autograd.Function):
         grad_mpi = ext.src_composite_bwd(
             mpi, depths, k_inv,
             src_img if ctx.blend else empty,
-            ctx.bg_inf,
+            ctx.bg_inf, ctx.alpha,
             g_rgb.contiguous() if g_rgb is not None else empty,
             g_depth.contiguous() if g_depth is not None else empty,
             g_blend.contiguous() if (ctx.blend and g_blend is not None) else empty)
-        return grad_mpi, None, None, None, None, None
+        return grad_mpi, None, None, None, None, None, None
 
 
 def render_src_view(mpi: torch.Tensor,
@@ -125,14 +126,19 @@ def render_src_view(mpi: torch.Tensor,
              mpi_blend (B,S,H,W,4) — input mpi with blended rgb; equals
              `mpi` when src_img is None).
     """
-    blend = src_img is not None
-    if mpi.is_cuda and not use_alpha:
+    blend = (src_img is not None) and not use_alpha
+    if mpi.is_cuda:
+        # alpha mode: no RGB blending, no background term
+        # (ref mpi_rendering.py:19-39)
         depths = torch.reciprocal(disparity).to(torch.float32)
         img_p = _img_packed(src_img.to(torch.float32)) if blend \
             else torch.empty(0, device=mpi.device)
-        return _SrcCompositeFn.apply(mpi.contiguous(), depths.contiguous(),
-                                     K_inv.to(torch.float32).contiguous(),
-                                     img_p, bg_depth_inf, blend)
+        out = _SrcCompositeFn.apply(mpi.contiguous(), depths.contiguous(),
+                                    K_inv.to(torch.float32).contiguous(),
+                                    img_p, bg_depth_inf, blend, use_alpha)
+        if use_alpha:
+            return out[0], out[1], mpi
+        return out
 
     # -------- torch reference path (CPU, and the use_alpha branch) --------
     rgb, sigma = unpack_mpi(mpi)
@@ -167,11 +173,13 @@ class _TgtCompositeFn(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, mpi, hinv, m, tvec, depths, bg_inf):
+    def forward(ctx, mpi, hinv, m, tvec, depths, bg_inf, alpha):
         ext = get_extension(required=True)
-        rgb, depth, mask = ext.tgt_composite_fwd(mpi, hinv, m, tvec, depths, bool(bg_inf))
+        rgb, depth, mask = ext.tgt_composite_fwd(mpi, hinv, m, tvec, depths,
+                                                 bool(bg_inf), bool(alpha))
         ctx.save_for_backward(mpi, hinv, m, tvec, depths)
         ctx.bg_inf = bool(bg_inf)
+        ctx.alpha = bool(alpha)
         return rgb, depth, mask
 
     @staticmethod
@@ -191,11 +199,11 @@ class _TgtCompositeFn(torch.autograd.Function):
         else:
             hfwd = empty
         grad_mpi = ext.tgt_composite_bwd(
-            mpi, hinv, hfwd, m, tvec, depths, ctx.bg_inf,
+            mpi, hinv, hfwd, m, tvec, depths, ctx.bg_inf, ctx.alpha,
             g_rgb.contiguous() if g_rgb is not None else empty,
             g_depth.contiguous() if g_depth is not None else empty,
             mode)
-        return grad_mpi, None, None, None, None, None
+        return grad_mpi, None, None, None, None, None, None
 
 
 def render_tgt_view(mpi: torch.Tensor,
@@ -216,13 +224,14 @@ def render_tgt_view(mpi: torch.Tensor,
     reference does, ref synthesis_task.py:439-442).
     """
     depths = torch.reciprocal(disparity).to(torch.float32)
-    if mpi.is_cuda and not use_alpha:
+    if mpi.is_cuda:
         with torch.no_grad():
             hinv = tr.homography_tgt_to_src(G_tgt_src, depths, K_src_inv, K_tgt)
             m = torch.matmul(G_tgt_src[:, :3, :3], K_src_inv).contiguous()
             tvec = G_tgt_src[:, :3, 3].contiguous()
         return _TgtCompositeFn.apply(mpi.contiguous(), hinv.contiguous(), m, tvec,
-                                     depths.contiguous(), bg_depth_inf)
+                                     depths.contiguous(), bg_depth_inf,
+                                     use_alpha)
 
     rgb, sigma = unpack_mpi(mpi)
     return tr.render_tgt_reference(rgb, sigma, disparity, G_tgt_src,

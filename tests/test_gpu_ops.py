@@ -545,9 +545,9 @@ def test_tgt_backward_gather_matches_scatter(bg_inf):
 
     empty = torch.empty(0, device="cuda:0", dtype=torch.float32)
     gm_scatter = ext.tgt_composite_bwd(mpi, hinv, empty, m, tvec, depths,
-                                       bg_inf, g_rgb, g_depth, 0)
+                                       bg_inf, False, g_rgb, g_depth, 0)
     gm_gather = ext.tgt_composite_bwd(mpi, hinv, hfwd, m, tvec, depths,
-                                      bg_inf, g_rgb, g_depth, 1)
+                                      bg_inf, False, g_rgb, g_depth, 1)
     torch.testing.assert_close(gm_gather, gm_scatter, rtol=1e-4, atol=1e-4)
     assert not torch.equal(gm_gather, torch.zeros_like(gm_gather))
 
@@ -647,3 +647,50 @@ def test_edge_aware_v2_fused_matches_eager():
         torch.testing.assert_close(loss.cpu(), loss_ref, rtol=1e-4, atol=1e-5)
         torch.testing.assert_close(d_gpu.grad.cpu(), d_ref.grad,
                                    rtol=1e-3, atol=1e-5)
+
+
+def test_alpha_composite_matches_oracle():
+    """use_alpha mode runs the FUSED kernels (round-1 weak 6: it used to
+    fall back to eager): fwd + bwd vs the fp64 torch oracle for both the
+    src composite and the novel-view render."""
+    from mine_amd.ops import torch_ref as trf
+    from mine_amd.ops.renderer import pack_mpi, render_src_view, render_tgt_view
+
+    rgb, sigma, disparity, K, K_inv, G, img = _mk_scene(S=12, H=24, W=32)
+    alpha = (torch.rand_like(sigma) * 0.85 + 0.05)
+    gseed = torch.Generator().manual_seed(3)
+    wr = torch.randn(rgb.shape[0], 3, rgb.shape[-2], rgb.shape[-1], generator=gseed)
+    wd = torch.randn(rgb.shape[0], 1, rgb.shape[-2], rgb.shape[-1], generator=gseed)
+
+    def run(device, dtype=torch.float32):
+        cast = lambda t: t.detach().clone().to(device=device, dtype=dtype)
+        r = cast(rgb).requires_grad_(True)
+        a = cast(alpha).requires_grad_(True)
+        if device == "cuda:0":
+            mpi = pack_mpi(r, a)
+            s_rgb, s_depth, _ = render_src_view(mpi, cast(disparity),
+                                                cast(K_inv), use_alpha=True)
+            t_rgb, t_depth, _ = render_tgt_view(mpi, cast(disparity), cast(G),
+                                                cast(K_inv), cast(K),
+                                                use_alpha=True)
+        else:
+            grid = trf.make_meshgrid(rgb.shape[-2], rgb.shape[-1])
+            xyz = trf.src_plane_xyz(grid, cast(disparity),
+                                    cast(K_inv)).to(dtype)
+            s_rgb, _ = trf.alpha_composite(a, r)
+            s_depth, _ = trf.alpha_composite(a, xyz[:, :, 2:])
+            t_rgb, t_depth, _ = trf.render_tgt_reference(
+                r, a, cast(disparity), cast(G), cast(K_inv), cast(K),
+                use_alpha=True)
+        loss = (s_rgb * cast(wr)).sum() + (s_depth * cast(wd)).sum() + \
+            (t_rgb * cast(wr)).sum() + (t_depth * cast(wd)).sum()
+        loss.backward()
+        return (s_rgb.detach().float().cpu(), t_rgb.detach().float().cpu(),
+                s_depth.detach().float().cpu(), t_depth.detach().float().cpu(),
+                r.grad.float().cpu(), a.grad.float().cpu())
+
+    g_gpu = run("cuda:0")
+    g_cpu = run("cpu", dtype=torch.float64)
+    for got, ref, tol in zip(g_gpu, g_cpu, (1e-4, 1e-3, 1e-3, 1e-2,
+                                            1e-3, 1e-3)):
+        torch.testing.assert_close(got, ref, rtol=1e-3, atol=tol)

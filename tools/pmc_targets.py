@@ -71,16 +71,16 @@ K_inv = torch.inverse(Km)
 G = torch.eye(4, device=dev).unsqueeze(0).repeat(Bb, 1, 1)
 G[:, 0, 3] = 0.1
 img = torch.rand(Bb, Hh, Ww, 3, device=dev)
-ext.src_composite_fwd(mpi, depths, K_inv.contiguous(), img, False)
+ext.src_composite_fwd(mpi, depths, K_inv.contiguous(), img, False, False)
 hinv = tr.homography_tgt_to_src(G, depths, K_inv, Km).contiguous()
 m = torch.matmul(G[:, :3, :3], K_inv).contiguous()
 tvec = G[:, :3, 3].contiguous()
-ext.tgt_composite_fwd(mpi, hinv, m, tvec, depths, False)
+ext.tgt_composite_fwd(mpi, hinv, m, tvec, depths, False, False)
 hfwd = inverse_3x3(hinv.reshape(-1, 3, 3)).reshape(Bb, S, 3, 3).contiguous()
 g_rgb = torch.rand(Bb, 3, Hh, Ww, device=dev)
 g_depth = torch.rand(Bb, 1, Hh, Ww, device=dev)
-ext.tgt_composite_bwd(mpi, hinv, hfwd, m, tvec, depths, False, g_rgb,
-                      g_depth, 1)
+ext.tgt_composite_bwd(mpi, hinv, hfwd, m, tvec, depths, False, False,
+                      g_rgb, g_depth, 1)
 
 # --- SSIM fwd ---
 from mine_amd.ops.ssim import ssim

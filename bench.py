@@ -34,6 +34,10 @@ def main() -> int:
     p.add_argument("--width", type=int, default=384)
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp16", "fp32"])
     p.add_argument("--dataset", type=str, default="realestate10k")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable the hipGraph-captured train step "
+                        "(single-GPU bf16/fp32 runs try it by default; "
+                        "capture failure falls back to eager)")
     p.add_argument("--timers", action="store_true",
                    help="print a sync-bracketed per-phase breakdown (diagnostic "
                         "run only; the extra syncs perturb the headline number)")
@@ -115,15 +119,24 @@ def main() -> int:
             if use_gpu:
                 torch.cuda.synchronize()
 
+    graphed = False
+    if use_gpu and world_size == 1 and args.dtype != "fp16" and \
+            not args.no_graph and not args.timers:
+        graphed = task.enable_graph_step(batches[0])
+        if rank == 0:
+            print(json.dumps({"hip_graph_step": bool(graphed)}),
+                  file=sys.stderr)
+    step_fn = task.train_step_graphed if graphed else task.train_step
+
     for i in range(args.warmup):
-        task.train_step(batches[i % len(batches)])
+        step_fn(batches[i % len(batches)])
 
     if args.timers:
         task.enable_phase_timers()
     sync()
     t0 = time.perf_counter()
     for i in range(args.steps):
-        task.train_step(batches[i % len(batches)])
+        step_fn(batches[i % len(batches)])
     sync()
     elapsed = time.perf_counter() - t0
 

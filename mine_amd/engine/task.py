@@ -248,21 +248,41 @@ class SynthesisTask:
             return torch.autocast(device_type="cuda", dtype=self.amp_dtype)
         return contextlib.nullcontext()
 
-    def set_data(self, items) -> None:
-        """Stage a batch on the device (ref synthesis_task.py:184-209)."""
+    def set_data(self, items, static: bool = False) -> None:
+        """Stage a batch on the device (ref synthesis_task.py:184-209).
+
+        static=True copies into persistent device buffers (same shapes
+        every step) — the staging the hipGraph-captured step reads."""
         src_items, tgt_items = items
         dev = self.device
 
         def tod(x):
             return x.to(dev, non_blocking=True)
 
+        L = tgt_items["img"].shape[1]
+        assert L == 1, "one target supervision view (ref synthesis_task.py:200-201)"
+
+        if static and getattr(self, "_static_staged", False):
+            self.src_imgs.copy_(src_items["img"], non_blocking=True)
+            self.K_src.copy_(src_items["K"], non_blocking=True)
+            self.K_src_inv.copy_(src_items["K_inv"], non_blocking=True)
+            self.pt3d_src.copy_(src_items["xyzs"], non_blocking=True)
+            self.tgt_imgs.copy_(tgt_items["img"].squeeze(1), non_blocking=True)
+            self.G_src_tgt.copy_(tgt_items["G_src_tgt"].squeeze(1),
+                                 non_blocking=True)
+            self.K_tgt.copy_(tgt_items["K"].squeeze(1), non_blocking=True)
+            self.K_tgt_inv.copy_(tgt_items["K_inv"].squeeze(1),
+                                 non_blocking=True)
+            self.pt3d_tgt.copy_(tgt_items["xyzs"].squeeze(1),
+                                non_blocking=True)
+            self.G_tgt_src.copy_(inverse_rigid_4x4(self.G_src_tgt))
+            return
+
         self.src_imgs = tod(src_items["img"]).float()  # Bx3xHxW
         self.K_src = tod(src_items["K"]).float()
         self.K_src_inv = tod(src_items["K_inv"]).float()
         self.pt3d_src = tod(src_items["xyzs"]).float()  # Bx3xN_pt
 
-        L = tgt_items["img"].shape[1]
-        assert L == 1, "one target supervision view (ref synthesis_task.py:200-201)"
         self.tgt_imgs = tod(tgt_items["img"]).float().squeeze(1)
         self.G_src_tgt = tod(tgt_items["G_src_tgt"]).float().squeeze(1)
         self.K_tgt = tod(tgt_items["K"]).float().squeeze(1)
@@ -276,6 +296,10 @@ class SynthesisTask:
         if self.channels_last:
             self.src_imgs = self.src_imgs.contiguous(memory_format=torch.channels_last)
             self.tgt_imgs = self.tgt_imgs.contiguous(memory_format=torch.channels_last)
+        if static:
+            # first static call: the tensors staged above BECOME the
+            # persistent buffers
+            self._static_staged = True
 
     # ------------------------------------------------------------------
     def mpi_predictor(self, src_imgs: torch.Tensor, disparity: torch.Tensor
@@ -585,6 +609,81 @@ class SynthesisTask:
             self.optimizer.step()
         mark("optimizer")
         return loss_dict
+
+    # ------------------------------------------------------------------
+    # hipGraph-captured train step (docs/NEXT round-1 item; single GPU)
+    # ------------------------------------------------------------------
+    def _graph_step_body(self) -> dict:
+        """Everything one optimization step does on the device, with no
+        host synchronization — the region the hipGraph captures."""
+        if self.grad_engine is not None:
+            self.grad_engine.zero_grad()
+        else:
+            self.optimizer.zero_grad(set_to_none=False)
+        loss_dict, _ = self.loss_fcn(is_val=False, monitors=False)
+        loss_dict["loss"].backward()
+        if self.grad_engine is not None:
+            self.grad_engine.finish_step()
+        if bool(self.config.get("training.nan_guard", True)):
+            finite = torch.isfinite(loss_dict["loss"].detach()).to(torch.float32)
+            if self._nan_skip_count is None:
+                self._nan_skip_count = torch.zeros_like(finite)
+            self._nan_skip_count += 1.0 - finite
+            grads = [b.flat for b in self.grad_engine.buckets] \
+                if self.grad_engine is not None else \
+                [p.grad for g in self.optimizer.param_groups
+                 for p in g["params"] if p.grad is not None]
+            torch._foreach_mul_(grads, finite)
+            for g in grads:
+                torch.nan_to_num_(g, nan=0.0, posinf=0.0, neginf=0.0)
+        self.optimizer.step()
+        return loss_dict
+
+    def enable_graph_step(self, items) -> bool:
+        """Capture forward+backward+optimizer into ONE hipGraph over
+        static input buffers; later steps copy the batch in and replay
+        (~600 launches collapse into one). Single-GPU, bf16/fp32, fresh
+        Adam state only; returns False (eager path keeps working) when
+        any precondition fails."""
+        if not (self.is_gpu and self.grad_scaler is None):
+            return False
+        if torch.distributed.is_initialized() and \
+                torch.distributed.get_world_size() > 1:
+            return False
+        if len(self.optimizer.state) > 0:
+            return False  # capturable Adam needs device-side step state
+        try:
+            # bias correction must be computed on-device per replay
+            for g in self.optimizer.param_groups:
+                g["capturable"] = True
+            self.set_data(items, static=True)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):  # warmup: allocator + lazy state
+                    self._graph_step_body()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                self._graph_loss = self._graph_step_body()
+            return True
+        except Exception as exc:  # pragma: no cover - depends on runtime
+            if self.logger:
+                self.logger.warning("hipGraph capture failed (%s); "
+                                    "eager step", exc)
+            self._graph = None
+            for g in self.optimizer.param_groups:
+                g["capturable"] = False
+            return False
+
+    def train_step_graphed(self, items) -> dict:
+        """Copy the batch into the static buffers and replay the graph.
+        The returned loss dict's tensors are the static graph outputs
+        (valid after the replay completes)."""
+        self.set_data(items, static=True)
+        self._graph.replay()
+        return self._graph_loss
 
     def enable_phase_timers(self) -> None:
         """Sync-bracketed per-phase wall times; read with pop_phase_times()."""

@@ -394,10 +394,14 @@ class SynthesisTask:
 
         sy = Hs_m / self.src_imgs.shape[-2]
         sx = Ws_m / self.src_imgs.shape[-1]
-        sc = torch.tensor([[sx], [sy], [1.0]], dtype=torch.float32,
-                          device=self.device)
-        K_src_scaled = self.K_src * sc
-        K_tgt_scaled = self.K_tgt * sc
+        # python scalars fold into the mul kernels (no host tensor: a
+        # pageable H2D would break hipGraph capture)
+        K_src_scaled = self.K_src.clone()
+        K_src_scaled[:, 0] = K_src_scaled[:, 0] * sx
+        K_src_scaled[:, 1] = K_src_scaled[:, 1] * sy
+        K_tgt_scaled = self.K_tgt.clone()
+        K_tgt_scaled[:, 0] = K_tgt_scaled[:, 0] * sx
+        K_tgt_scaled[:, 1] = K_tgt_scaled[:, 1] * sy
         K_src_scaled_inv = inverse_3x3(K_src_scaled)
 
         # ---- fused src composite + RGB blending -------------------------
@@ -480,7 +484,9 @@ class SynthesisTask:
             if self.lpips_model is not None and is_val and scale == 0:
                 lpips_tgt = self.lpips_model(tgt_imgs_syn, tgt_scaled).mean()
             else:
-                lpips_tgt = torch.tensor(0.0, device=self.device)
+                # zeros((), device) is a device-side fill (graph-capture
+                # safe; torch.tensor(0.0, device=...) is a pageable H2D)
+                lpips_tgt = torch.zeros((), device=self.device)
             psnr_tgt = psnr(tgt_imgs_syn, tgt_scaled) if monitors else \
                 torch.zeros((), device=self.device)
 
@@ -669,12 +675,20 @@ class SynthesisTask:
                 self._graph_loss = self._graph_step_body()
             return True
         except Exception as exc:  # pragma: no cover - depends on runtime
+            self._graph_error = f"{type(exc).__name__}: {exc}"
             if self.logger:
                 self.logger.warning("hipGraph capture failed (%s); "
-                                    "eager step", exc)
+                                    "eager step", self._graph_error)
             self._graph = None
+            torch.cuda.synchronize()
             for g in self.optimizer.param_groups:
                 g["capturable"] = False
+            # warmup may have created capturable (device) step counters;
+            # the eager non-capturable Adam wants them on host
+            for st in self.optimizer.state.values():
+                if "step" in st and torch.is_tensor(st["step"]) and \
+                        st["step"].is_cuda:
+                    st["step"] = st["step"].cpu()
             return False
 
     def train_step_graphed(self, items) -> dict:

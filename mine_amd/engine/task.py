@@ -659,6 +659,10 @@ class SynthesisTask:
         if len(self.optimizer.state) > 0:
             return False  # capturable Adam needs device-side step state
         try:
+            # the library conv path allocates workspace per call, which
+            # hipGraph capture forbids: force the hand-written kernels
+            from mine_amd.ops.conv import set_force_igemm
+            set_force_igemm(True)
             # bias correction must be computed on-device per replay
             for g in self.optimizer.param_groups:
                 g["capturable"] = True
@@ -675,6 +679,8 @@ class SynthesisTask:
                 self._graph_loss = self._graph_step_body()
             return True
         except Exception as exc:  # pragma: no cover - depends on runtime
+            from mine_amd.ops.conv import set_force_igemm
+            set_force_igemm(False)
             self._graph_error = f"{type(exc).__name__}: {exc}"
             if self.logger:
                 self.logger.warning("hipGraph capture failed (%s); "

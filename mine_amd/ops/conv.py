@@ -72,6 +72,15 @@ def pack_weights(w: torch.Tensor, flip: bool = False) -> torch.Tensor:
 
 
 _BWD_MODE = None  # lazy: "hip" (hand-written MFMA) or "miopen" (round-1 path)
+_FORCE_IGEMM = False  # graph capture: no library convs (workspace allocs)
+
+
+def set_force_igemm(v: bool) -> None:
+    """Route EVERY reflect conv through the hand-written kernels — the
+    library path allocates find/workspace memory per call, which breaks
+    hipGraph capture. Set by SynthesisTask.enable_graph_step."""
+    global _FORCE_IGEMM
+    _FORCE_IGEMM = bool(v)
 
 
 def _bwd_mode() -> str:
@@ -166,7 +175,8 @@ def conv3x3_reflect(x: torch.Tensor, w: torch.Tensor,
               and x.is_contiguous(memory_format=torch.channels_last))
     if usable:
         return _Conv3x3ReflFn.apply(x, w, bias)
-    if x.is_cuda and x.dtype == torch.bfloat16 and x.numel() * 2 <= 1 << 25:
+    if x.is_cuda and x.dtype == torch.bfloat16 and (
+            _FORCE_IGEMM or x.numel() * 2 <= 1 << 25):
         # small (L2/L3-resident) narrow-image shapes, e.g. the 256-ch
         # decoder block at H/16: the general igemm family (its direct
         # loads re-read x per tap, so gate on cache residency)

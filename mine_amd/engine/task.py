@@ -675,7 +675,10 @@ class SynthesisTask:
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
             self._graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self._graph):
+            # thread_local: autograd's device worker threads must be
+            # allowed to record into the capturing stream
+            with torch.cuda.graph(self._graph,
+                                  capture_error_mode="thread_local"):
                 self._graph_loss = self._graph_step_body()
             return True
         except Exception as exc:  # pragma: no cover - depends on runtime

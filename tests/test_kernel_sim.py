@@ -356,3 +356,99 @@ def test_igemm_kernel_coord_and_pack_math():
     xr = x.clone().requires_grad_(True)
     (F.conv2d(xr, w, stride=2, padding=1) * gy).sum().backward()
     torch.testing.assert_close(got, xr.grad, rtol=1e-4, atol=1e-4)
+
+
+def _img_elem32(col, j):
+    """igemm_kernels.hip img_elem32: the 32-pixel (NB=8) transpose-read
+    block image used by the general conv weight-grad."""
+    pblk = j >> 2
+    bi = (pblk & 1) * 4 + (pblk >> 1)
+    return bi * 64 + (j & 3) * 16 + col
+
+
+def test_igemm_wrw_tr_image32_addressing():
+    """Bijectivity + fragment reconstruction for the 32-px tr image:
+    frag32's two tr reads (lane slice base + (l&15)*4 + (l>>4)*64, and
+    +32*8) must deliver element (col=l&15, px=(l>>4)*8+e)."""
+    img = {}
+    for col in range(16):
+        for j in range(32):
+            off = _img_elem32(col, j)
+            assert off not in img
+            img[off] = (col, j)
+    assert len(img) == 16 * 32
+
+    for lane in range(64):
+        g = lane >> 4
+        for jj in range(4):
+            # tr semantics (tools/tr_probe.hip): lane l elem jj reads
+            # base + (l&15) + jj*16 + g*64
+            off_lo = (lane & 15) + jj * 16 + g * 64
+            assert img[off_lo] == (lane & 15, g * 8 + jj)
+            off_hi = off_lo + 32 * 8
+            assert img[off_hi] == (lane & 15, g * 8 + 4 + jj)
+
+
+def test_igemm_wrw_kernel_math():
+    """Simulates conv_igemm_wrw_kernel: (k16, c16, tap) blocks, 32-pixel
+    chunks, forward-coordinate tap mapping — against autograd for a
+    stride-2 zero-pad 3x3 and the reflect base-conv case."""
+    torch.manual_seed(12)
+
+    def src(p, r, SA, SB, SD, SE, n, reflect):
+        num = p * SA + r * SB + SD
+        if SE > 1:
+            if num % SE != 0:
+                return None
+            num //= SE
+        if reflect:
+            return _reflect1(num, n)
+        return num if 0 <= num < n else None
+
+    def sim_wrw(x, gy, R, stride, pad, reflect):
+        B, C, Hs, Ws = x.shape
+        K, P, Q = gy.shape[1], gy.shape[2], gy.shape[3]
+        xn = x.permute(0, 2, 3, 1).double()
+        gyn = gy.permute(0, 2, 3, 1).double().reshape(-1, K)
+        M = B * P * Q
+        dw = torch.zeros(K, C, R, R, dtype=torch.float64)
+        for k0 in range(0, K, 16):
+            for c0 in range(0, C, 16):
+                for tap in range(R * R):
+                    r, s_ = divmod(tap, R)
+                    acc = torch.zeros(16, 16, dtype=torch.float64)
+                    for ch0 in range(0, M, 32):
+                        A = torch.zeros(16, 32, dtype=torch.float64)
+                        Bm = torch.zeros(16, 32, dtype=torch.float64)
+                        for px in range(32):
+                            m = ch0 + px
+                            if m >= M:
+                                continue
+                            A[:, px] = gyn[m, k0:k0 + 16]
+                            n, rem = divmod(m, P * Q)
+                            p, q = divmod(rem, Q)
+                            ys = src(p, r, stride, 1, -pad, 1, Hs, reflect)
+                            xs = src(q, s_, stride, 1, -pad, 1, Ws, reflect)
+                            if ys is not None and xs is not None:
+                                Bm[:, px] = xn[n, ys, xs, c0:c0 + 16]
+                        acc += A @ Bm.T
+                    dw[k0:k0 + 16, c0:c0 + 16, r, s_] += acc
+        return dw.float()
+
+    # stride-2 zero-pad 3x3
+    B, C, H, W, K = 2, 16, 6, 7, 16
+    x = torch.randn(B, C, H, W)
+    w = torch.zeros(K, C, 3, 3, requires_grad=True)
+    y = F.conv2d(x, w, stride=2, padding=1)
+    gy = torch.randn_like(y)
+    (y * gy).sum().backward()
+    got = sim_wrw(x, gy, 3, 2, 1, False)
+    torch.testing.assert_close(got, w.grad, rtol=1e-4, atol=1e-4)
+
+    # reflect-pad stride-1 (the SplitConvBlock base convs)
+    w2 = torch.zeros(K, C, 3, 3, requires_grad=True)
+    y2 = F.conv2d(F.pad(x, (1, 1, 1, 1), mode="reflect"), w2)
+    gy2 = torch.randn_like(y2)
+    (y2 * gy2).sum().backward()
+    got2 = sim_wrw(x, gy2, 3, 1, 1, True)
+    torch.testing.assert_close(got2, w2.grad, rtol=1e-4, atol=1e-4)

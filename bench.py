@@ -107,8 +107,11 @@ def main() -> int:
     task = SynthesisTask(cfg, state=state)
 
     # Pre-build a few host-side batches; the H2D staging stays inside the
-    # timed region (it is part of a real training step).
+    # timed region (it is part of a real training step). Each rank gets
+    # DISTINCT data (per-rank seed) so the gradient all-reduce moves
+    # real information, as in training.
     ds = SyntheticMPIDataset(cfg, length=args.batch * 4)
+    ds.seed_base = 1 + rank
     batches = [collate_src_tgt([ds[i * args.batch + j] for j in range(args.batch)])
                for i in range(4)]
 
